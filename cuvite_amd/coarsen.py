@@ -1,0 +1,159 @@
+"""Graph coarsening (phase transition): collapse communities into vertices.
+
+Reference: distReNumber + fill_newEdgesMap + send_newEdges
+(rebuild.cpp:27-454). MI355X-native redesign: the reference renumbers through
+std::map and aggregates edges in nested maps on the host, serially; here every
+step is a device tensor op (sort / unique / searchsorted / segment sums), and
+edges are pre-aggregated locally before the exchange so the all-to-all traffic
+is the coarse edge list, not the fine one.
+
+Steps:
+  1. Surviving communities are routed to their owners (owner of community c =
+     owner of vertex id c); each owner assigns dense new ids to its own
+     survivors; a count allgather turns them into global new ids (contiguous
+     in rank order, so the new 1-D partition is exactly "what each owner
+     numbered"), matching distReNumber's offsets (rebuild.cpp:181).
+  2. Every rank resolves old-comm -> new-id for all communities it references
+     (its own labels + its ghosts' labels) with one owner lookup.
+  3. Edge tuples (new_src, new_dst, w) are aggregated locally, routed to the
+     new owner of new_src, merged again, and assembled into the new CSR.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Tuple
+
+import torch
+
+from .graph import DistGraph, Graph, Partition
+from .halo import build_halo, exchange_ghost_labels
+from .parallel import Comm
+
+
+def _owner_lookup(comm: Comm, part: Partition, query: torch.Tensor,
+                  reply_fn) -> torch.Tensor:
+    """Generic owner-based lookup: route sorted `query` gids to owners, each
+    owner answers via reply_fn(local_ids) -> values, results return aligned
+    with `query` (which must be sorted)."""
+    dev = query.device
+    if comm.world == 1:
+        return reply_fn(query)
+    parts = part.parts.to(dev)
+    offs = torch.searchsorted(query, parts)
+    reqs = [query[offs[p]:offs[p + 1]] for p in range(comm.world)]
+    got = comm.all_to_all_v(reqs)
+    replies = [reply_fn(g) if g.numel() else
+               torch.empty(0, dtype=torch.int64, device=dev) for g in got]
+    back = comm.all_to_all_v(replies, recv_counts=[int(r.numel()) for r in reqs])
+    return torch.cat([back[p] for p in range(comm.world)])
+
+
+def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor
+            ) -> Tuple[DistGraph, Callable[[torch.Tensor], torch.Tensor]]:
+    """Build the next-level graph from community labels `cvect` (int64 [nv],
+    global old-community ids). Returns (new DistGraph for this rank, renum)
+    where renum maps any tensor of old community gids to new vertex gids
+    (collective: all ranks must call it together with their own queries)."""
+    dev = dg.g.device
+    base, bound = dg.base, dg.bound
+    world, rank = comm.world, comm.rank
+
+    # --- 1. survivors to owners; owners number their own ---------------------
+    my_labels = torch.unique(cvect)  # sorted
+    if world == 1:
+        survivors = my_labels
+    else:
+        parts = dg.partition.parts.to(dev)
+        offs = torch.searchsorted(my_labels, parts)
+        reqs = [my_labels[offs[p]:offs[p + 1]] for p in range(world)]
+        got = comm.all_to_all_v(reqs)
+        survivors = torch.unique(torch.cat(got))  # sorted, all owned by me
+    lnc = survivors.numel()
+    counts = torch.zeros(world, dtype=torch.int64, device=comm.device)
+    counts[rank] = lnc
+    counts = comm.allgather_counts(counts).sum(dim=0) if world > 1 else counts
+    new_offsets = torch.zeros(world + 1, dtype=torch.int64)
+    new_offsets[1:] = torch.cumsum(counts.cpu(), dim=0)
+    my_off = int(new_offsets[rank])
+    gnc = int(new_offsets[-1])
+    assert gnc < (1 << 31), "coarse graph must have < 2^31 communities"
+    new_part = Partition.from_bounds(new_offsets)
+
+    def reply_newid(gids: torch.Tensor) -> torch.Tensor:
+        return my_off + torch.searchsorted(survivors, gids)
+
+    def renum(gids: torch.Tensor) -> torch.Tensor:
+        flat_sorted, order = torch.sort(gids)
+        vals = _owner_lookup(comm, dg.partition, flat_sorted, reply_newid)
+        out = torch.empty_like(vals)
+        out[order] = vals
+        return out
+
+    # --- 2. resolve new ids for all referenced communities -------------------
+    halo = build_halo(dg, comm)
+    ghost_cvect = exchange_ghost_labels(halo, cvect)
+    all_cvect = torch.cat([cvect, ghost_cvect])
+    ref_uniq = torch.unique(all_cvect)  # sorted
+    ref_new = _owner_lookup(comm, dg.partition, ref_uniq, reply_newid)
+    # map labels via position in ref_uniq
+    new_of_all = ref_new[torch.searchsorted(ref_uniq, all_cvect)]
+
+    # --- 3. coarse edges: aggregate locally, route, merge, CSR ---------------
+    seg = torch.repeat_interleave(torch.arange(dg.nv, device=dev),
+                                  dg.g.degrees())
+    s_new = new_of_all[seg]
+    t_new = new_of_all[halo.tails_dense.to(torch.int64)]
+    w = dg.g.weights
+    s_agg, t_agg, w_agg = _aggregate(s_new, t_new, w, gnc)
+
+    if world > 1:
+        npdev = new_part.parts.to(dev)
+        offs = torch.searchsorted(s_agg, npdev)
+        sp = [s_agg[offs[p]:offs[p + 1]] for p in range(world)]
+        got_s = comm.all_to_all_v(sp)
+        cnts = [int(g.numel()) for g in got_s]
+        got_t = comm.all_to_all_v([t_agg[offs[p]:offs[p + 1]] for p in range(world)],
+                                  recv_counts=cnts)
+        got_w = comm.all_to_all_v([w_agg[offs[p]:offs[p + 1]] for p in range(world)],
+                                  recv_counts=cnts)
+        s_agg = torch.cat(got_s)
+        t_agg = torch.cat(got_t)
+        w_agg = torch.cat(got_w)
+        s_agg, t_agg, w_agg = _aggregate(s_agg, t_agg, w_agg, gnc)
+
+    nbase = new_part.base(rank)
+    nv_new = new_part.nv_local(rank)
+    rowptr = torch.zeros(nv_new + 1, dtype=torch.int64, device=dev)
+    if s_agg.numel():
+        rowptr[1:] = torch.cumsum(
+            torch.bincount(s_agg - nbase, minlength=nv_new), dim=0)
+    new_g = Graph(rowptr, t_agg, w_agg)
+    return DistGraph(new_g, new_part, rank), renum
+
+
+def _aggregate(s: torch.Tensor, t: torch.Tensor, w: torch.Tensor, gnc: int):
+    """Merge duplicate (s, t) directed edges, summing weights; returns sorted
+    by (s, t). Keys fit int64 because gnc < 2^31."""
+    key = s * gnc + t
+    key_s, order = torch.sort(key)
+    w_s = w[order]
+    uniq, inv = torch.unique_consecutive(key_s, return_inverse=True)
+    w_out = torch.zeros(uniq.numel(), dtype=w.dtype, device=w.device)
+    w_out.index_add_(0, inv, w_s)
+    return uniq // gnc, uniq % gnc, w_out
+
+
+def remap_labels(dg: DistGraph, comm: Comm, assign: torch.Tensor,
+                 cvect: torch.Tensor) -> torch.Tensor:
+    """Compose clusterings across a phase: `assign[j]` is the current-level
+    vertex gid that original vertex j belongs to; return cvect[assign[j]]
+    fetched from the owning ranks (distributed version of the reference's
+    root-side commAll[p] = cvectAll[commAll[p]], main.cpp:397-404)."""
+    def reply(gids: torch.Tensor) -> torch.Tensor:
+        return cvect[gids - dg.base]
+
+    flat_sorted, order = torch.sort(assign)
+    vals = _owner_lookup(comm, dg.partition, flat_sorted, reply)
+    out = torch.empty_like(vals)
+    out[order] = vals
+    return out

@@ -1,0 +1,272 @@
+"""In-memory graph generators: R-MAT (benchmark headline), RGG (reference
+parity), the reproducible parallel LCG, and the Zachary karate-club test graph.
+
+Reference analogs: generateInMemGraph/generateRGG (distgraph.cpp:341-933) and
+the LCG parallel-prefix RNG (utils.hpp:76-271). Differences (deliberate,
+MI355X-first):
+  - The LCG stream is jumped ahead with modular exponentiation instead of the
+    MPI 2x2 matrix parallel prefix: rank-independent, communication-free, and
+    the SAME stream for any process count.
+  - RGG vertex coordinates are pure functions of the global vertex id, so the
+    generated graph is identical for any P (the reference's strip-local
+    streams make the graph depend on P).
+  - R-MAT (Graph500-style) is the headline generator (BASELINE.json configs);
+    edges are generated in fixed global chunks seeded with counter-based
+    Philox, so the edge list is identical for any P.
+"""
+
+from __future__ import annotations
+
+import math
+
+import numpy as np
+import torch
+
+from .graph import Graph, DistGraph, Partition
+
+# ---------------------------------------------------------------- LCG ------
+
+MLCG = 2147483647  # 2^31 - 1
+ALCG = 16807       # 7^5 (Park-Miller MINSTD), ref utils.hpp:25-27
+
+
+def _seed_x0(seed: int) -> int:
+    """Derive x0 from a 32-bit seed (ref reseeder uses std::seed_seq; we use
+    splitmix32 — documented deviation, stream is still P-independent)."""
+    z = (seed + 0x9E3779B9) & 0xFFFFFFFF
+    z = ((z ^ (z >> 16)) * 0x85EBCA6B) & 0xFFFFFFFF
+    z = ((z ^ (z >> 13)) * 0xC2B2AE35) & 0xFFFFFFFF
+    z = z ^ (z >> 16)
+    x0 = z % MLCG
+    return x0 if x0 != 0 else 1
+
+
+class LCG:
+    """Park-Miller LCG with O(log k) jump-ahead: element k of the global
+    stream is x0 * ALCG^k mod MLCG. Any rank can materialize any slice of the
+    one global sequence without communication."""
+
+    def __init__(self, seed: int = 1):
+        self.x0 = _seed_x0(seed)
+
+    def slice(self, start: int, count: int) -> np.ndarray:
+        """Stream elements [start, start+count) as int64 in [0, MLCG)."""
+        if count <= 0:
+            return np.zeros(0, dtype=np.int64)
+        a_start = pow(ALCG, start, MLCG)
+        first = (self.x0 * a_start) % MLCG
+        # powers table A^0..A^(count-1) by doubling (log2 numpy passes)
+        pows = np.ones(1, dtype=np.int64)
+        while pows.size < count:
+            a_len = pow(ALCG, int(pows.size), MLCG)
+            pows = np.concatenate([pows, (pows * a_len) % MLCG])
+        pows = pows[:count]
+        return (first * pows) % MLCG  # both < 2^31 so product < 2^62: no overflow
+
+    def uniform(self, start: int, count: int) -> np.ndarray:
+        """Stream elements mapped to [0,1) doubles (ref generate(), mult =
+        1/MLCG)."""
+        return self.slice(start, count).astype(np.float64) * (1.0 / MLCG)
+
+
+# --------------------------------------------------------------- R-MAT -----
+
+RMAT_A, RMAT_B, RMAT_C = 0.57, 0.19, 0.19  # Graph500 defaults (d = 0.05)
+_CHUNK = 1 << 22  # global generation chunk (4M edges): P-independent unit
+
+
+def _rmat_chunk(scale: int, n_edges: int, seed: int, chunk_id: int) -> tuple:
+    """Generate `n_edges` undirected R-MAT edge tuples (u, v, w) for global
+    chunk `chunk_id`, deterministically from (seed, chunk_id)."""
+    rng = np.random.Generator(
+        np.random.Philox(key=[(seed << 32) | 0xC0FFEE, chunk_id]))
+    u = np.zeros(n_edges, dtype=np.int64)
+    v = np.zeros(n_edges, dtype=np.int64)
+    ab = RMAT_A + RMAT_B
+    a_norm = RMAT_A / ab
+    c_norm = RMAT_C / (1.0 - ab)
+    for _ in range(scale):
+        r1 = rng.random(n_edges)
+        r2 = rng.random(n_edges)
+        ubit = r1 > ab
+        vbit = np.where(ubit, r2 > c_norm, r2 > a_norm)
+        u = (u << 1) | ubit
+        v = (v << 1) | vbit
+    w = rng.random(n_edges)
+    return u, v, w
+
+
+def rmat_edges(scale: int, edgefactor: int, seed: int,
+               part_lo: float = 0.0, part_hi: float = 1.0):
+    """Undirected R-MAT edge tuples for the fraction [part_lo, part_hi) of the
+    global edge list, chunk-deterministic: the union over disjoint fractions
+    covering [0,1) equals the P=1 edge list exactly."""
+    ne = edgefactor << scale
+    nchunks = (ne + _CHUNK - 1) // _CHUNK
+    c_lo = int(math.floor(part_lo * nchunks))
+    c_hi = int(math.floor(part_hi * nchunks)) if part_hi < 1.0 else nchunks
+    us, vs, ws = [], [], []
+    for c in range(c_lo, c_hi):
+        n_e = min(_CHUNK, ne - c * _CHUNK)
+        u, v, w = _rmat_chunk(scale, n_e, seed, c)
+        us.append(u)
+        vs.append(v)
+        ws.append(w)
+    if not us:
+        z = np.zeros(0, dtype=np.int64)
+        return z, z, np.zeros(0, dtype=np.float64)
+    return np.concatenate(us), np.concatenate(vs), np.concatenate(ws)
+
+
+def rmat_graph(scale: int, edgefactor: int = 16, seed: int = 1,
+               weight_dtype: torch.dtype = torch.float64) -> Graph:
+    """Whole symmetrized R-MAT graph on one process (tests / small runs)."""
+    u, v, w = rmat_edges(scale, edgefactor, seed)
+    nv = 1 << scale
+    src = torch.from_numpy(np.concatenate([u, v]))
+    dst = torch.from_numpy(np.concatenate([v, u]))
+    ww = torch.from_numpy(np.concatenate([w, w])).to(weight_dtype)
+    return Graph.from_edge_tuples(nv, src, dst, ww)
+
+
+# ----------------------------------------------------------------- RGG -----
+
+def rgg_radius(nv: int) -> float:
+    """Reference cutoff: mean of the connectivity threshold sqrt(ln nv/(pi nv))
+    and sqrt(2.0736/nv) (distgraph.cpp:344-349)."""
+    rc = math.sqrt(math.log(nv) / (math.pi * nv))
+    rt = math.sqrt(2.0736 / nv)
+    return 0.5 * (rc + rt)
+
+
+def _rgg_coords(lcg: LCG, lo: int, hi: int, nv: int):
+    """Coordinates of global vertices [lo, hi): x = u_{2i} in [0,1),
+    y = (i + u_{2i+1})/nv — ids are ordered along y, so contiguous id ranges
+    are y-bands and any rank can regenerate any window without communication."""
+    u = lcg.uniform(2 * lo, 2 * (hi - lo))
+    x = u[0::2]
+    y = (np.arange(lo, hi, dtype=np.float64) + u[1::2]) / nv
+    return x, y
+
+
+def rgg_local_edges(nv: int, rank: int, nranks: int, seed: int = 1,
+                    random_edge_percent: float = 0.0):
+    """Directed edge tuples (src global, dst global, weight) for all src owned
+    by `rank` under the contiguous partition. Euclidean-distance weights
+    (distgraph.cpp:361-362); optional extra random edges with U(0.01, 1.0)
+    weights (distgraph.cpp:757)."""
+    part = Partition.contiguous(nv, nranks)
+    base, bound = part.base(rank), part.bound(rank)
+    rn = rgg_radius(nv)
+    lcg = LCG(seed)
+    halo = int(math.ceil(rn * nv)) + 1
+    lo, hi = max(0, base - halo), min(nv, bound + halo)
+    x, y = _rgg_coords(lcg, lo, hi, nv)
+    ids = np.arange(lo, hi, dtype=np.int64)
+
+    # spatial hash on cells of size rn; candidate pairs from 3x3 neighborhoods
+    ncell = max(1, int(1.0 / rn))
+    cx = np.minimum((x * ncell).astype(np.int64), ncell - 1)
+    cy = np.minimum((y * ncell).astype(np.int64), ncell - 1)
+    cell = cy * ncell + cx
+    order = np.argsort(cell, kind="stable")
+    cell_s = cell[order]
+    starts = np.searchsorted(cell_s, np.arange(ncell * ncell))
+    ends = np.searchsorted(cell_s, np.arange(ncell * ncell), side="right")
+
+    is_mine = (ids >= base) & (ids < bound)
+    srcs, dsts, wts = [], [], []
+    mine_idx = np.nonzero(is_mine)[0]
+    # gather candidates per 3x3 neighborhood of each cell containing my points
+    my_cells = np.unique(cell[mine_idx])
+    for c in my_cells:
+        cyy, cxx = divmod(int(c), ncell)
+        cand = []
+        for dy in (-1, 0, 1):
+            for dx in (-1, 0, 1):
+                yy, xx = cyy + dy, cxx + dx
+                if 0 <= yy < ncell and 0 <= xx < ncell:
+                    c2 = yy * ncell + xx
+                    cand.append(order[starts[c2]:ends[c2]])
+        cand = np.concatenate(cand) if cand else np.zeros(0, dtype=np.int64)
+        pts = order[starts[int(c)]:ends[int(c)]]
+        pts = pts[is_mine[pts]]
+        if pts.size == 0 or cand.size == 0:
+            continue
+        dxm = x[pts][:, None] - x[cand][None, :]
+        dym = y[pts][:, None] - y[cand][None, :]
+        d = np.sqrt(dxm * dxm + dym * dym)
+        pi, ci = np.nonzero((d <= rn) & (ids[pts][:, None] != ids[cand][None, :]))
+        srcs.append(ids[pts][pi])
+        dsts.append(ids[cand][ci])
+        wts.append(d[pi, ci])
+
+    if srcs:
+        src = np.concatenate(srcs)
+        dst = np.concatenate(dsts)
+        w = np.concatenate(wts)
+    else:
+        src = np.zeros(0, dtype=np.int64)
+        dst = np.zeros(0, dtype=np.int64)
+        w = np.zeros(0, dtype=np.float64)
+
+    if random_edge_percent > 0.0:
+        # reference: ceil(percent * ne_global / 100) extra undirected random
+        # edges (distgraph.cpp:706-760). P-independent: both endpoints drawn
+        # from a dedicated Philox stream; each rank emits the directed copies
+        # whose src it owns.
+        ne_base_local = src.size
+        # estimate global count deterministically: use expected edges via a
+        # fixed chunked draw keyed only by (seed, nv)
+        rng = np.random.Generator(
+            np.random.Philox(key=[(seed << 32) | 0xFACADE, nv & 0x7FFFFFFF]))
+        n_rand = int(math.ceil(random_edge_percent * nv / 100.0))
+        ru = rng.integers(0, nv, n_rand)
+        rv = rng.integers(0, nv, n_rand)
+        rw = rng.uniform(0.01, 1.0, n_rand)
+        for a, b in ((ru, rv), (rv, ru)):
+            m = (a >= base) & (a < bound) & (a != b)
+            src = np.concatenate([src, a[m]])
+            dst = np.concatenate([dst, b[m]])
+            w = np.concatenate([w, rw[m]])
+
+    return src, dst, w
+
+
+def rgg_dist_graph(nv: int, rank: int, nranks: int, seed: int = 1,
+                   random_edge_percent: float = 0.0,
+                   weight_dtype: torch.dtype = torch.float64) -> DistGraph:
+    src, dst, w = rgg_local_edges(nv, rank, nranks, seed, random_edge_percent)
+    part = Partition.contiguous(nv, nranks)
+    g = Graph.from_edge_tuples(part.nv_local(rank), torch.from_numpy(src),
+                               torch.from_numpy(dst),
+                               torch.from_numpy(w).to(weight_dtype),
+                               base=part.base(rank))
+    return DistGraph(g, part, rank)
+
+
+# ------------------------------------------------------------- karate ------
+
+_KARATE_EDGES = [
+    (0, 1), (0, 2), (0, 3), (0, 4), (0, 5), (0, 6), (0, 7), (0, 8), (0, 10),
+    (0, 11), (0, 12), (0, 13), (0, 17), (0, 19), (0, 21), (0, 31), (1, 2),
+    (1, 3), (1, 7), (1, 13), (1, 17), (1, 19), (1, 21), (1, 30), (2, 3),
+    (2, 7), (2, 8), (2, 9), (2, 13), (2, 27), (2, 28), (2, 32), (3, 7),
+    (3, 12), (3, 13), (4, 6), (4, 10), (5, 6), (5, 10), (5, 16), (6, 16),
+    (8, 30), (8, 32), (8, 33), (9, 33), (13, 33), (14, 32), (14, 33),
+    (15, 32), (15, 33), (18, 32), (18, 33), (19, 33), (20, 32), (20, 33),
+    (22, 32), (22, 33), (23, 25), (23, 27), (23, 29), (23, 32), (23, 33),
+    (24, 25), (24, 27), (24, 31), (25, 31), (26, 29), (26, 33), (27, 33),
+    (28, 31), (28, 33), (29, 32), (29, 33), (30, 32), (30, 33), (31, 32),
+    (31, 33), (32, 33),
+]
+
+
+def karate_graph(weight_dtype: torch.dtype = torch.float64) -> Graph:
+    """Zachary karate club (34 vertices, 78 undirected edges, unit weights) —
+    the reference's canonical smoke-test graph (README:60)."""
+    e = torch.tensor(_KARATE_EDGES, dtype=torch.int64)
+    src = torch.cat([e[:, 0], e[:, 1]])
+    dst = torch.cat([e[:, 1], e[:, 0]])
+    w = torch.ones(src.numel(), dtype=weight_dtype)
+    return Graph.from_edge_tuples(34, src, dst, w)

@@ -1,0 +1,391 @@
+"""Multi-phase distributed Louvain orchestration.
+
+Structure mirrors the reference: an outer phase loop
+(local-moving until convergence, then graph coarsening) around an inner
+iteration loop (halo fill -> local move -> community update -> remote update
+-> modularity -> convergence test). Variants: plain, early-termination
+(freeze / probabilistic, ref louvain.cpp:7-424), coloring-ordered and
+vertex-ordered moves (ref louvain.cpp:756-2101), threshold cycling
+(main.cpp:225-239).
+
+All state lives on the compute device; per-iteration host work is zero on the
+GPU path (the reference rebuilds its dense community remap on the host every
+iteration, louvain_cuda.cu:2260-2378 — here it is torch sort/unique/
+searchsorted on device).
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+
+from .graph import DistGraph, Partition
+from .halo import (HaloContext, build_halo, exchange_ghost_labels,
+                   fetch_remote_comm_info, push_remote_deltas)
+from .local_move import MoveInputs, local_move_torch, modularity_parts
+from .parallel import Comm
+
+TERMINATION_PHASE_COUNT = 200   # ref utils.hpp:17-19
+MAX_TOTAL_ITERS = 10000         # ref main.cpp:486-494
+ET_CUTOFF = 90                  # ref louvain.hpp:76 (absolute count)
+P_CUTOFF = 0.02                 # ref louvain.hpp:78
+
+
+@dataclass
+class LouvainConfig:
+    threshold: float = 1.0e-6
+    threshold_scaling: bool = False      # -i
+    one_phase: bool = False              # -p
+    early_term: int = 0                  # -t 1..4 (0 = off)
+    et_delta: float = 1.0                # -a (alpha for -t 2/4)
+    coloring: bool = False               # -c
+    ordering: bool = False               # -d
+    max_colors: int = 8                  # -c/-d arg (nHash*2)
+    backend: str = "auto"                # "auto" | "torch" | "hip"
+    max_phases: int = TERMINATION_PHASE_COUNT
+    max_iters_per_phase: int = 10000
+    verbose: bool = False
+
+
+@dataclass
+class LouvainResult:
+    modularity: float
+    communities: torch.Tensor   # final community id per ORIGINAL local vertex
+    phases: int
+    total_iters: int
+    times: dict = field(default_factory=dict)
+    modularity_per_level: list = field(default_factory=list)
+
+
+def _threshold_for_phase(cfg: LouvainConfig, short_phase: int) -> float:
+    """Threshold cycling (ref main.cpp:225-239)."""
+    if not cfg.threshold_scaling or cfg.one_phase:
+        return 1.0e-6
+    sp = short_phase % 13
+    if sp <= 2:
+        return 1.0e-3
+    if sp <= 6:
+        return 1.0e-4
+    if sp <= 9:
+        return 1.0e-5
+    return 1.0e-6
+
+
+def _pick_move_fn(cfg: LouvainConfig, device: torch.device):
+    if cfg.backend == "torch":
+        return local_move_torch
+    if cfg.backend in ("auto", "hip") and device.type == "cuda":
+        from . import ops
+        return ops.local_move  # raises if the HIP extension is missing
+    if cfg.backend == "hip":
+        raise RuntimeError("backend='hip' requires a CUDA/HIP device")
+    return local_move_torch
+
+
+class PhaseState:
+    """Mutable per-phase state for one rank (all tensors on dg's device)."""
+
+    def __init__(self, dg: DistGraph, comm: Comm):
+        dev = dg.g.device
+        self.dg = dg
+        self.comm = comm
+        self.halo = build_halo(dg, comm)
+        nv = dg.nv
+        W = dg.g.weights.dtype
+        self.v_degree = dg.local_degree_sum()                       # W [nv]
+        # community aggregates, indexed by (gid - base) for owned comms
+        self.local_size = torch.ones(nv, dtype=torch.int64, device=dev)
+        self.local_degree = self.v_degree.clone()
+        # global labels (ref distInitComm: singletons)
+        arange = torch.arange(dg.base, dg.bound, device=dev)
+        self.curr_comm = arange.clone()
+        self.past_comm = arange.clone()
+        # constant = 1/(2m) (ref distCalcConstantForSecondTerm)
+        tw = float(self.v_degree.to(torch.float64).sum())
+        tw = comm.allreduce_scalar(tw)
+        self.constant = 1.0 / tw
+        # ET state
+        self.active = torch.ones(nv, dtype=torch.bool, device=dev)
+        self.stable_count = torch.zeros(nv, dtype=torch.int16, device=dev)
+        self.move_prob = torch.ones(nv, dtype=torch.float32, device=dev)
+
+    def densify(self, ghost_comm: torch.Tensor):
+        """Map global labels to the per-iteration dense community space.
+        Returns (curr_dense [nv+ng] int32, remote_gids sorted, comm_size,
+        comm_degree, comm_gid) with remote info fetched from owners."""
+        dg, dev = self.dg, self.dg.g.device
+        base, bound, nv = dg.base, dg.bound, dg.nv
+        all_labels = torch.cat([self.curr_comm, ghost_comm])
+        is_local = (all_labels >= base) & (all_labels < bound)
+        remote_gids = torch.unique(all_labels[~is_local])
+        r_size, r_degree = fetch_remote_comm_info(
+            self.halo, remote_gids, self.local_size, self.local_degree)
+        dense = torch.empty_like(all_labels)
+        dense[is_local] = all_labels[is_local] - base
+        if remote_gids.numel():
+            dense[~is_local] = nv + torch.searchsorted(remote_gids,
+                                                       all_labels[~is_local])
+        comm_size = torch.cat([self.local_size, r_size])
+        comm_degree = torch.cat([self.local_degree, r_degree])
+        comm_gid = torch.cat([torch.arange(base, bound, device=dev), remote_gids])
+        return dense.to(torch.int32), remote_gids, comm_size, comm_degree, comm_gid
+
+    def apply_moves(self, target_gid: torch.Tensor, remote_gids: torch.Tensor):
+        """Apply community size/degree deltas for vertices that moved
+        (ref 4-case update, louvain.cpp:2308-2376 + updateRemoteCommunities)."""
+        dg = self.dg
+        base, bound = dg.base, dg.bound
+        moved = target_gid != self.curr_comm
+        if not bool(moved.any()):
+            # still must participate in the collective delta push
+            if self.comm.world > 1:
+                empty_i = torch.empty(0, dtype=torch.int64, device=target_gid.device)
+                empty_w = torch.empty(0, dtype=self.local_degree.dtype,
+                                      device=target_gid.device)
+                push_remote_deltas(self.halo, empty_i, empty_i, empty_w,
+                                   self.local_size, self.local_degree)
+            return
+        src = self.curr_comm[moved]
+        dst = target_gid[moved]
+        vdeg = self.v_degree[moved]
+        gids = torch.cat([src, dst])
+        dsize = torch.cat([-torch.ones_like(src), torch.ones_like(dst)])
+        ddeg = torch.cat([-vdeg, vdeg])
+        is_local = (gids >= base) & (gids < bound)
+        li = gids[is_local] - base
+        self.local_size.index_add_(0, li, dsize[is_local])
+        self.local_degree.index_add_(0, li, ddeg[is_local])
+        if self.comm.world > 1:
+            push_remote_deltas(self.halo, gids[~is_local], dsize[~is_local],
+                               ddeg[~is_local], self.local_size, self.local_degree)
+        else:
+            assert bool(is_local.all())
+
+
+def _et_update(state: PhaseState, cfg: LouvainConfig, target: torch.Tensor,
+               rng: torch.Generator) -> Optional[int]:
+    """Early-termination bookkeeping after an iteration. Returns the global
+    frozen count for -t 3/4 (ref louvain.cpp:115-120, 172-182, 378-395)."""
+    if cfg.early_term in (1, 3):
+        stable = (target == state.curr_comm) & (state.curr_comm == state.past_comm)
+        state.stable_count = torch.where(stable, state.stable_count + 1,
+                                         torch.zeros_like(state.stable_count))
+        state.active &= state.stable_count < 3
+    elif cfg.early_term in (2, 4):
+        moved = target != state.curr_comm
+        state.move_prob = torch.where(
+            moved, torch.ones_like(state.move_prob),
+            state.move_prob * (1.0 - cfg.et_delta))
+        draw = torch.rand(state.move_prob.shape, generator=rng,
+                          device="cpu").to(state.move_prob.device)
+        state.active = (state.move_prob >= P_CUTOFF) & \
+            ((draw < state.move_prob) | moved)
+    if cfg.early_term in (3, 4):
+        frozen = int((~state.active).sum())
+        return int(state.comm.allreduce_scalar(float(frozen)))
+    return None
+
+
+def run_phase(dg: DistGraph, comm: Comm, cfg: LouvainConfig,
+              lower: float, threshold: float,
+              colors: Optional[torch.Tensor] = None,
+              num_colors: int = 0):
+    """One Louvain phase (local moving to convergence). Returns
+    (prev_mod, cvect global labels [nv], iters).
+    Ref: distLouvainMethod (louvain.cpp:425-588) and the coloring/ordering
+    variants (louvain.cpp:756-2101)."""
+    state = PhaseState(dg, comm)
+    move_fn = _pick_move_fn(cfg, dg.g.device)
+    prev_mod = lower
+    iters = 0
+    rng = torch.Generator().manual_seed(12345 + comm.rank)
+
+    use_colors = colors is not None and num_colors > 0
+    color_order: List[torch.Tensor] = []
+    if use_colors:
+        for c in range(num_colors):
+            color_order.append(torch.nonzero(colors == c, as_tuple=True)[0])
+
+    while iters < cfg.max_iters_per_phase:
+        iters += 1
+        target = _one_sweep(state, cfg, move_fn,
+                            color_order if use_colors and cfg.coloring else None)
+
+        # modularity over the whole sweep
+        curr_mod = _modularity(state)
+
+        frozen = _et_update(state, cfg, target, rng)
+
+        # rotate BEFORE convergence check only in the reference order:
+        # check first, rotate after (ref louvain.cpp:541-567)
+        if (curr_mod - prev_mod) < threshold:
+            break
+        prev_mod = max(curr_mod, lower)
+        state.past_comm, state.curr_comm = state.curr_comm, target
+
+        if frozen is not None and frozen >= ET_CUTOFF:
+            break
+
+    cvect = state.past_comm
+    return prev_mod, cvect, iters
+
+
+def _one_sweep(state: PhaseState, cfg: LouvainConfig, move_fn,
+               color_order: Optional[List[torch.Tensor]]):
+    """One local-moving sweep over all vertices. Plain mode: one simultaneous
+    move computation. Coloring mode (-c): sequential passes over color
+    classes, with a fresh halo exchange and community update per color
+    (ref louvain.cpp:862-901)."""
+    nv = state.dg.nv
+    dev = state.dg.g.device
+
+    if color_order is None:
+        ghost_comm = exchange_ghost_labels(state.halo, state.curr_comm)
+        dense, remote_gids, c_size, c_degree, c_gid = state.densify(ghost_comm)
+        inp = MoveInputs(state.dg.g.rowptr, state.halo.tails_dense,
+                         state.dg.g.weights, dense, state.v_degree,
+                         c_size, c_degree, c_gid, state.constant)
+        tgt_dense, cw = move_fn(inp)
+        target = _dense_to_gid(state, tgt_dense, remote_gids)
+        if cfg.early_term:
+            target = torch.where(state.active, target, state.curr_comm)
+        state.cluster_weight = torch.where(
+            state.active, cw, torch.zeros(1, dtype=cw.dtype, device=dev)) \
+            if cfg.early_term else cw
+        state.apply_moves(target, remote_gids)
+        return target
+
+    # coloring-ordered: process color classes sequentially, adopting moves
+    # after each class (labels change within the sweep)
+    work = state.curr_comm.clone()
+    cw_total = torch.zeros(nv, dtype=state.dg.g.weights.dtype, device=dev)
+    for vidx in color_order:
+        ghost_comm = exchange_ghost_labels(state.halo, work)
+        saved = state.curr_comm
+        state.curr_comm = work
+        dense, remote_gids, c_size, c_degree, c_gid = state.densify(ghost_comm)
+        state.curr_comm = saved
+        inp = MoveInputs(state.dg.g.rowptr, state.halo.tails_dense,
+                         state.dg.g.weights, dense, state.v_degree,
+                         c_size, c_degree, c_gid, state.constant)
+        tgt_dense, cw = move_fn(inp)
+        tgt = _dense_to_gid_from(work, tgt_dense, remote_gids,
+                                 state.dg.base, state.dg.bound)
+        mask = torch.zeros(nv, dtype=torch.bool, device=dev)
+        mask[vidx] = True
+        if cfg.early_term:
+            mask &= state.active
+        tgt = torch.where(mask, tgt, work)
+        cw_total = torch.where(mask, cw, cw_total)
+        # apply deltas for this color class only
+        saved_curr = state.curr_comm
+        state.curr_comm = work
+        state.apply_moves(tgt, remote_gids)
+        state.curr_comm = saved_curr
+        work = tgt
+    state.cluster_weight = cw_total
+    return work
+
+
+def _dense_to_gid(state: PhaseState, dense: torch.Tensor,
+                  remote_gids: torch.Tensor) -> torch.Tensor:
+    return _dense_to_gid_from(state.curr_comm, dense, remote_gids,
+                              state.dg.base, state.dg.bound)
+
+
+def _dense_to_gid_from(curr: torch.Tensor, dense: torch.Tensor,
+                       remote_gids: torch.Tensor, base: int, bound: int):
+    nv = curr.numel()
+    d = dense.to(torch.int64)
+    out = torch.where(d < nv, d + base,
+                      remote_gids[(d - nv).clamp(min=0, max=max(remote_gids.numel() - 1, 0))]
+                      if remote_gids.numel() else d + base)
+    return out
+
+
+def _modularity(state: PhaseState) -> float:
+    parts = modularity_parts(state.cluster_weight, state.local_degree)
+    state.comm.allreduce_sum_(parts)
+    c = state.constant
+    return float(parts[0]) * c - float(parts[1]) * c * c
+
+
+def louvain(dg: DistGraph, comm: Optional[Comm] = None,
+            cfg: Optional[LouvainConfig] = None) -> LouvainResult:
+    """Full multi-phase Louvain (ref main.cpp:218-495). Returns the final
+    community id per ORIGINAL local vertex (contiguous global ids) and the
+    final modularity."""
+    from .coarsen import coarsen, remap_labels
+    from .coloring import distance1_coloring
+
+    comm = comm or Comm(dg.g.device)
+    cfg = cfg or LouvainConfig()
+    dev = dg.g.device
+
+    t_start = time.perf_counter()
+    orig_assign = torch.arange(dg.base, dg.bound, device=dev)  # current label of my originals
+    prev_mod, curr_mod = -1.0, -1.0
+    phase = 0
+    short_phase = 0
+    tot_iters = 0
+    mods = []
+    level = dg
+    times = {"coloring": 0.0, "clustering": 0.0, "rebuild": 0.0}
+
+    while True:
+        threshold = _threshold_for_phase(cfg, short_phase) \
+            if cfg.threshold_scaling else cfg.threshold
+
+        colors = None
+        num_colors = 0
+        if (cfg.coloring or cfg.ordering) and phase == 0:
+            t0 = time.perf_counter()
+            colors, num_colors = distance1_coloring(level, comm,
+                                                    n_hash=max(1, cfg.max_colors // 2))
+            times["coloring"] += time.perf_counter() - t0
+
+        t0 = time.perf_counter()
+        curr_mod, cvect, iters = run_phase(
+            level, comm, cfg, curr_mod, threshold,
+            colors=colors if cfg.coloring else None, num_colors=num_colors)
+        times["clustering"] += time.perf_counter() - t0
+        tot_iters += iters
+
+        if (curr_mod - prev_mod) > threshold:
+            # compose: originals currently assigned to level vertices; level
+            # vertex v now belongs to community cvect[v]
+            orig_assign = remap_labels(level, comm, orig_assign, cvect)
+            mods.append(curr_mod)
+            if cfg.one_phase:
+                break
+            t0 = time.perf_counter()
+            level, renum = coarsen(level, comm, cvect)
+            # cvect-composed orig_assign holds OLD comm gids; renumber them
+            orig_assign = renum(orig_assign)
+            times["rebuild"] += time.perf_counter() - t0
+        else:
+            if cfg.threshold_scaling and not cfg.one_phase and phase < 10:
+                curr_mod2, cvect, iters = run_phase(level, comm, cfg, curr_mod,
+                                                    1.0e-6)
+                tot_iters += iters
+                if (curr_mod2 - curr_mod) > 1.0e-6:
+                    orig_assign = remap_labels(level, comm, orig_assign, cvect)
+                    mods.append(curr_mod2)
+                    curr_mod = curr_mod2
+            break
+
+        prev_mod = curr_mod
+        curr_mod = -1.0
+        phase += 1
+        if cfg.threshold_scaling:
+            short_phase += 1
+        if phase >= cfg.max_phases or tot_iters > MAX_TOTAL_ITERS:
+            break
+
+    times["total"] = time.perf_counter() - t_start
+    final_mod = mods[-1] if mods else max(prev_mod, curr_mod)
+    return LouvainResult(final_mod, orig_assign, phase + 1, tot_iters,
+                         times, mods)

@@ -1,0 +1,80 @@
+import numpy as np
+import pytest
+import torch
+
+from cuvite_amd.graph import Graph, DistGraph, Partition, single_partition
+from cuvite_amd.generators import karate_graph, rmat_graph, LCG, rmat_edges
+
+
+def test_partition_contiguous():
+    p = Partition.contiguous(10, 3)
+    assert p.parts.tolist() == [0, 3, 6, 10]
+    assert p.owner(torch.tensor([0, 2, 3, 5, 6, 9])).tolist() == [0, 0, 1, 1, 2, 2]
+    assert p.nv_local(0) == 3 and p.nv_local(2) == 4
+
+
+def test_partition_edge_balanced():
+    # 4 vertices, degrees 10, 1, 1, 10
+    index = torch.tensor([0, 10, 11, 12, 22], dtype=torch.int64)
+    p = Partition.edge_balanced(index, 2)
+    # rank 0 should get vertex 0 (10 edges), rank 1 the rest (12)
+    assert p.parts[0] == 0 and p.parts[-1] == 4
+    e0 = int(index[p.parts[1]]) - int(index[p.parts[0]])
+    e1 = int(index[p.parts[2]]) - int(index[p.parts[1]])
+    assert abs(e0 - e1) <= 10
+
+
+def test_karate_csr():
+    g = karate_graph()
+    assert g.nv == 34
+    assert g.ne == 156  # 78 undirected edges, both directions
+    assert int(g.degrees().sum()) == 156
+    # symmetric
+    dg = single_partition(g)
+    assert dg.ghost_vertices().numel() == 0
+    vdeg = dg.local_degree_sum()
+    assert float(vdeg.sum()) == 156.0
+    assert float(vdeg[33]) == 17.0 and float(vdeg[0]) == 16.0
+
+
+def test_from_edge_tuples_sorted():
+    src = torch.tensor([2, 0, 1, 0], dtype=torch.int64)
+    dst = torch.tensor([1, 2, 0, 1], dtype=torch.int64)
+    w = torch.tensor([1.0, 2.0, 3.0, 4.0], dtype=torch.float64)
+    g = Graph.from_edge_tuples(3, src, dst, w)
+    assert g.rowptr.tolist() == [0, 2, 3, 4]
+    assert g.tails.tolist() == [1, 2, 0, 1]
+    assert g.weights.tolist() == [4.0, 2.0, 3.0, 1.0]
+
+
+def test_lcg_p_independent():
+    lcg = LCG(42)
+    full = lcg.slice(0, 100)
+    a = lcg.slice(0, 37)
+    b = lcg.slice(37, 63)
+    assert np.array_equal(np.concatenate([a, b]), full)
+    # recurrence holds
+    assert (full[1:] == (full[:-1] * 16807) % 2147483647).all()
+    u = lcg.uniform(0, 100)
+    assert (u >= 0).all() and (u < 1).all()
+
+
+def test_rmat_chunks_p_independent():
+    u0, v0, w0 = rmat_edges(10, 16, seed=7)
+    u1a, v1a, w1a = rmat_edges(10, 16, seed=7, part_lo=0.0, part_hi=0.5)
+    u1b, v1b, w1b = rmat_edges(10, 16, seed=7, part_lo=0.5, part_hi=1.0)
+    assert np.array_equal(np.concatenate([u1a, u1b]), u0)
+    assert np.array_equal(np.concatenate([v1a, v1b]), v0)
+    assert np.array_equal(np.concatenate([w1a, w1b]), w0)
+    assert u0.size == 16 * 1024
+    assert u0.max() < 1024 and v0.max() < 1024
+
+
+def test_rmat_graph_symmetric():
+    g = rmat_graph(8, 8, seed=3)
+    assert g.nv == 256
+    assert g.ne == 2 * 8 * 256
+    # undirected: total weight of (u,v) equals (v,u)
+    dg = single_partition(g)
+    vdeg = dg.local_degree_sum()
+    assert float(vdeg.sum()) == pytest.approx(2 * float(g.weights.sum()) / 2, rel=1e-12)

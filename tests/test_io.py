@@ -1,0 +1,72 @@
+import os
+
+import pytest
+import torch
+
+from cuvite_amd.generators import karate_graph, rmat_graph
+from cuvite_amd.io import (load_dist_graph, load_graph, write_graph,
+                           read_header, load_ground_truth, write_communities)
+
+
+def test_roundtrip_whole(tmp_path):
+    g = karate_graph()
+    p = str(tmp_path / "karate.bin")
+    write_graph(p, g)
+    assert read_header(p) == (34, 156)
+    g2 = load_graph(p)
+    assert torch.equal(g.rowptr, g2.rowptr)
+    assert torch.equal(g.tails, g2.tails)
+    assert torch.allclose(g.weights, g2.weights)
+
+
+def test_sharded_load_matches_whole(tmp_path):
+    g = rmat_graph(7, 8, seed=5)
+    p = str(tmp_path / "rmat.bin")
+    write_graph(p, g)
+    for nranks in (2, 3, 4):
+        off = 0
+        for r in range(nranks):
+            dg = load_dist_graph(p, r, nranks)
+            nvl = dg.nv
+            assert torch.equal(dg.g.rowptr + int(g.rowptr[off]),
+                               g.rowptr[off:off + nvl + 1])
+            e0, e1 = int(g.rowptr[off]), int(g.rowptr[off + nvl])
+            assert torch.equal(dg.g.tails, g.tails[e0:e1])
+            off += nvl
+        assert off == g.nv
+
+
+def test_balanced_load(tmp_path):
+    g = rmat_graph(7, 8, seed=5)
+    p = str(tmp_path / "rmat.bin")
+    write_graph(p, g)
+    shards = [load_dist_graph(p, r, 4, balanced=True) for r in range(4)]
+    assert sum(s.nv for s in shards) == g.nv
+    assert sum(s.ne for s in shards) == g.ne
+    ne_max = max(s.ne for s in shards)
+    # balanced shards should be closer to ne/4 than the worst-case skew
+    assert ne_max <= g.ne  # sanity
+    tails = torch.cat([s.g.tails for s in shards])
+    assert torch.equal(tails, g.tails)
+
+
+def test_unit_weights(tmp_path):
+    g = karate_graph()
+    p = str(tmp_path / "k.bin")
+    write_graph(p, g)
+    dg = load_dist_graph(p, 0, 1, unit_weights=True)
+    assert float(dg.g.weights.sum()) == g.ne
+
+
+def test_ground_truth_io(tmp_path):
+    p = str(tmp_path / "gt.txt")
+    comm = torch.tensor([0, 0, 1, 1, 2], dtype=torch.int64)
+    write_communities(p, comm)
+    gt = load_ground_truth(p)
+    assert torch.equal(gt, comm)
+    # 1-based variant
+    with open(p, "w") as f:
+        for v in range(5):
+            f.write(f"{v+1} {int(comm[v])+1}\n")
+    gt1 = load_ground_truth(p, zero_based=False)
+    assert torch.equal(gt1, comm)

@@ -1,0 +1,114 @@
+import numpy as np
+import pytest
+import torch
+
+from cuvite_amd.generators import karate_graph, rmat_graph
+from cuvite_amd.graph import single_partition
+from cuvite_amd.local_move import (MoveInputs, local_move_pydict,
+                                   local_move_torch, modularity_parts)
+
+
+def _singleton_inputs(g):
+    dg = single_partition(g)
+    vdeg = dg.local_degree_sum()
+    nv = g.nv
+    return MoveInputs(
+        rowptr=g.rowptr,
+        tails=g.tails.to(torch.int32),
+        weights=g.weights,
+        curr_comm=torch.arange(nv, dtype=torch.int32),
+        v_degree=vdeg,
+        comm_size=torch.ones(nv, dtype=torch.int64),
+        comm_degree=vdeg.clone(),
+        comm_gid=torch.arange(nv, dtype=torch.int64),
+        constant=1.0 / float(vdeg.to(torch.float64).sum()),
+    )
+
+
+def _random_comm_inputs(g, seed):
+    rng = torch.Generator().manual_seed(seed)
+    nv = g.nv
+    dg = single_partition(g)
+    vdeg = dg.local_degree_sum()
+    curr = torch.randint(0, nv, (nv,), generator=rng)
+    # aggregates consistent with the labels
+    size = torch.zeros(nv, dtype=torch.int64)
+    size.index_add_(0, curr, torch.ones(nv, dtype=torch.int64))
+    cdeg = torch.zeros(nv, dtype=vdeg.dtype)
+    cdeg.index_add_(0, curr, vdeg)
+    return MoveInputs(g.rowptr, g.tails.to(torch.int32), g.weights,
+                      curr.to(torch.int32), vdeg, size, cdeg,
+                      torch.arange(nv, dtype=torch.int64),
+                      1.0 / float(vdeg.to(torch.float64).sum()))
+
+
+@pytest.mark.parametrize("case", ["karate_singleton", "karate_random",
+                                  "rmat_singleton", "rmat_random"])
+def test_torch_matches_pydict(case):
+    if case.startswith("karate"):
+        g = karate_graph()
+    else:
+        g = rmat_graph(7, 8, seed=11)
+    if case.endswith("singleton"):
+        inp = _singleton_inputs(g)
+    else:
+        inp = _random_comm_inputs(g, seed=23)
+    t_ref, cw_ref = local_move_pydict(inp)
+    t_vec, cw_vec = local_move_torch(inp)
+    assert torch.equal(t_ref.to(torch.int64), t_vec.to(torch.int64))
+    assert torch.allclose(cw_ref, cw_vec, atol=1e-12)
+
+
+def test_first_iteration_karate_moves():
+    g = karate_graph()
+    inp = _singleton_inputs(g)
+    target, cw = local_move_torch(inp)
+    # from singletons, no weight to own community yet
+    assert float(cw.sum()) == 0.0
+    # singleton guard: vertices only move to LOWER ids on the first sweep
+    t = target.to(torch.int64)
+    moved = t != torch.arange(34)
+    assert bool(moved.any())
+    assert bool((t[moved] < torch.arange(34)[moved]).all())
+
+
+def test_isolated_vertex_stays():
+    rowptr = torch.tensor([0, 2, 3, 3, 4], dtype=torch.int64)
+    tails = torch.tensor([1, 3, 0, 0], dtype=torch.int32)
+    w = torch.ones(4, dtype=torch.float64)
+    inp = MoveInputs(rowptr, tails, w,
+                     torch.arange(4, dtype=torch.int32),
+                     torch.tensor([2.0, 1.0, 0.0, 1.0], dtype=torch.float64),
+                     torch.ones(4, dtype=torch.int64),
+                     torch.tensor([2.0, 1.0, 0.0, 1.0], dtype=torch.float64),
+                     torch.arange(4, dtype=torch.int64),
+                     0.25)
+    t, cw = local_move_torch(inp)
+    assert int(t[2]) == 2
+    tr, cwr = local_move_pydict(inp)
+    assert torch.equal(t.to(torch.int64), tr.to(torch.int64))
+
+
+def test_self_loop_semantics():
+    # vertex 0 has a self loop; it contributes to clusterWeight but not to eix
+    rowptr = torch.tensor([0, 2, 3], dtype=torch.int64)
+    tails = torch.tensor([0, 1, 0], dtype=torch.int32)
+    w = torch.tensor([5.0, 1.0, 1.0], dtype=torch.float64)
+    vdeg = torch.tensor([6.0, 1.0], dtype=torch.float64)
+    inp = MoveInputs(rowptr, tails, w,
+                     torch.arange(2, dtype=torch.int32), vdeg,
+                     torch.ones(2, dtype=torch.int64), vdeg.clone(),
+                     torch.arange(2, dtype=torch.int64),
+                     1.0 / 7.0)
+    t, cw = local_move_torch(inp)
+    tr, cwr = local_move_pydict(inp)
+    assert torch.equal(t.to(torch.int64), tr.to(torch.int64))
+    assert torch.allclose(cw, cwr)
+    assert float(cw[0]) == 5.0  # self-loop counted in counter[cc]
+
+
+def test_modularity_parts():
+    cw = torch.tensor([1.0, 2.0], dtype=torch.float64)
+    cd = torch.tensor([3.0, 4.0], dtype=torch.float64)
+    p = modularity_parts(cw, cd)
+    assert p.tolist() == [3.0, 25.0]
