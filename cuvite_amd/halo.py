@@ -117,10 +117,12 @@ def fetch_remote_comm_info(ctx: HaloContext, remote_gids: torch.Tensor,
     comm, dg = ctx.comm, ctx.dg
     dev = remote_gids.device
     W = local_degree.dtype
-    if comm.world == 1 or remote_gids.numel() == 0:
+    if comm.world == 1:
         assert remote_gids.numel() == 0, "remote comms with world=1"
         return (torch.empty(0, dtype=torch.int64, device=dev),
                 torch.empty(0, dtype=W, device=dev))
+    # NOTE: even with no remote references of our own we must participate —
+    # peers may be requesting OUR community info (collective protocol).
     parts = dg.partition.parts.to(dev)
     offs = torch.searchsorted(remote_gids, parts)
     reqs = [remote_gids[offs[p]:offs[p + 1]] for p in range(comm.world)]
