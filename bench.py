@@ -1,0 +1,162 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Louvain local-moving iterations on an R-MAT graph.
+
+Metric (BASELINE.json): Louvain edges/sec (TEPS: global directed edge count x
+timed iterations / elapsed, the reference's definition, main.cpp:448,509) plus
+the final modularity, on R-MAT scale-26 with random edge weights, at 1-8
+MI355X GPUs (strong scaling: the graph is fixed, ranks each own a 1-D slice).
+
+One "step" = one full distributed local-moving iteration: ghost-label exchange
+(RCCL p2p) + remote community-info fetch + HIP local-move kernel + community
+delta push + modularity allreduce. Nothing is skipped inside the timed region.
+
+Launch (the driver does this):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from cuvite_amd.generators import rmat_edges_torch, rmat_edges  # noqa: E402
+from cuvite_amd.graph import Graph, DistGraph, Partition  # noqa: E402
+from cuvite_amd.louvain import (LouvainConfig, PhaseState, _modularity,
+                                _one_sweep, _pick_move_fn)  # noqa: E402
+from cuvite_amd.parallel import Comm, init_from_env  # noqa: E402
+
+
+def build_rmat_dist(scale, edgefactor, seed, comm, device, wdtype):
+    """Generate this rank's R-MAT slice on device and route directed edges to
+    their 1-D owners."""
+    nv = 1 << scale
+    part = Partition.contiguous(nv, comm.world)
+    lo = comm.rank / comm.world
+    hi = (comm.rank + 1) / comm.world
+    if device.type == "cuda":
+        u, v, w = rmat_edges_torch(scale, edgefactor, seed, lo, hi, device,
+                                   weight_dtype=wdtype)
+    else:
+        import numpy as np
+        uu, vv, ww = rmat_edges(scale, edgefactor, seed, lo, hi)
+        u = torch.from_numpy(uu)
+        v = torch.from_numpy(vv)
+        w = torch.from_numpy(ww).to(wdtype)
+    # symmetrize: directed copies
+    src = torch.cat([u, v])
+    dst = torch.cat([v, u])
+    ww = torch.cat([w, w])
+    del u, v, w
+    if comm.world > 1:
+        parts_dev = part.parts.to(device)
+        order = torch.argsort(src)
+        src, dst, ww = src[order], dst[order], ww[order]
+        del order
+        offs = torch.searchsorted(src, parts_dev)
+        sp_s = [src[offs[p]:offs[p + 1]] for p in range(comm.world)]
+        got_s = comm.all_to_all_v(sp_s)
+        cnts = [int(g.numel()) for g in got_s]
+        got_d = comm.all_to_all_v(
+            [dst[offs[p]:offs[p + 1]] for p in range(comm.world)], cnts)
+        got_w = comm.all_to_all_v(
+            [ww[offs[p]:offs[p + 1]] for p in range(comm.world)], cnts)
+        src = torch.cat(got_s)
+        dst = torch.cat(got_d)
+        ww = torch.cat(got_w)
+    g = Graph.from_edge_tuples(part.nv_local(comm.rank), src, dst, ww,
+                               base=part.base(comm.rank))
+    return DistGraph(g, part, comm.rank)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--scale", type=int, default=26)
+    ap.add_argument("--edgefactor", type=int, default=16)
+    ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--dtype", choices=["fp64", "fp32"], default="fp64")
+    ap.add_argument("--backend", default="auto",
+                    help="auto|hip|torch (torch = eager fallback, CPU test only)")
+    args = ap.parse_args()
+
+    comm = init_from_env()
+    device = comm.device
+    wdtype = torch.float64 if args.dtype == "fp64" else torch.float32
+
+    t_gen0 = time.perf_counter()
+    dg = build_rmat_dist(args.scale, args.edgefactor, args.seed, comm, device,
+                         wdtype)
+    ne_global = float(comm.allreduce_scalar(float(dg.ne)))
+    t_gen = time.perf_counter() - t_gen0
+
+    cfg = LouvainConfig(backend=args.backend)
+    state = PhaseState(dg, comm)
+    state.use_hip = device.type == "cuda" and args.backend in ("auto", "hip")
+    move_fn = _pick_move_fn(cfg, device)
+
+    def step():
+        target = _one_sweep(state, cfg, move_fn, None)
+        q = _modularity(state)
+        state.past_comm, state.curr_comm = state.curr_comm, target
+        return q
+
+    for _ in range(args.warmup):
+        q = step()
+
+    comm.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        q = step()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    comm.barrier()
+    elapsed = time.perf_counter() - t0
+    # max over ranks
+    if comm.world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t[0])
+
+    teps = ne_global * args.steps / elapsed
+    if comm.rank == 0:
+        out = {
+            "metric": "louvain_edges_per_sec",
+            "value": teps,
+            "unit": "edges/s",
+            "n_gpus": comm.world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": f"louvain-rmat-s{args.scale}-ef{args.edgefactor}",
+                "scale": args.scale,
+                "edgefactor": args.edgefactor,
+                "nv": 1 << args.scale,
+                "ne_directed": int(ne_global),
+                "modularity_last_step": q,
+                "gen_seconds": round(t_gen, 2),
+                "parallelism": f"graph1d-p{comm.world}",
+                "backend": args.backend,
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -118,6 +118,44 @@ def rmat_edges(scale: int, edgefactor: int, seed: int,
     return np.concatenate(us), np.concatenate(vs), np.concatenate(ws)
 
 
+def rmat_edges_torch(scale: int, edgefactor: int, seed: int,
+                     part_lo: float, part_hi: float, device,
+                     weight_dtype: torch.dtype = torch.float64):
+    """GPU-resident R-MAT edge generation, chunk-deterministic like
+    rmat_edges() but using torch's Philox on `device` (one seeded generator
+    per global chunk, so the graph is identical for any process count on the
+    same device type). Returns (u, v, w) undirected tuples on `device`."""
+    ne = edgefactor << scale
+    nchunks = (ne + _CHUNK - 1) // _CHUNK
+    c_lo = int(math.floor(part_lo * nchunks))
+    c_hi = int(math.floor(part_hi * nchunks)) if part_hi < 1.0 else nchunks
+    ab = RMAT_A + RMAT_B
+    a_norm = RMAT_A / ab
+    c_norm = RMAT_C / (1.0 - ab)
+    us, vs, ws = [], [], []
+    gen = torch.Generator(device=device)
+    for c in range(c_lo, c_hi):
+        n_e = min(_CHUNK, ne - c * _CHUNK)
+        gen.manual_seed((seed * 0x1F123BB5 + c) & 0x7FFFFFFFFFFF)
+        u = torch.zeros(n_e, dtype=torch.int64, device=device)
+        v = torch.zeros(n_e, dtype=torch.int64, device=device)
+        for _ in range(scale):
+            r1 = torch.rand(n_e, generator=gen, device=device)
+            r2 = torch.rand(n_e, generator=gen, device=device)
+            ubit = r1 > ab
+            vbit = torch.where(ubit, r2 > c_norm, r2 > a_norm)
+            u = (u << 1) | ubit
+            v = (v << 1) | vbit
+        w = torch.rand(n_e, generator=gen, device=device, dtype=weight_dtype)
+        us.append(u)
+        vs.append(v)
+        ws.append(w)
+    if not us:
+        z = torch.zeros(0, dtype=torch.int64, device=device)
+        return z, z, torch.zeros(0, dtype=weight_dtype, device=device)
+    return torch.cat(us), torch.cat(vs), torch.cat(ws)
+
+
 def rmat_graph(scale: int, edgefactor: int = 16, seed: int = 1,
                weight_dtype: torch.dtype = torch.float64) -> Graph:
     """Whole symmetrized R-MAT graph on one process (tests / small runs)."""

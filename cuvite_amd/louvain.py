@@ -199,6 +199,8 @@ def run_phase(dg: DistGraph, comm: Comm, cfg: LouvainConfig,
     variants (louvain.cpp:756-2101)."""
     state = PhaseState(dg, comm)
     move_fn = _pick_move_fn(cfg, dg.g.device)
+    state.use_hip = (dg.g.device.type == "cuda"
+                     and cfg.backend in ("auto", "hip"))
     prev_mod = lower
     iters = 0
     rng = torch.Generator().manual_seed(12345 + comm.rank)
@@ -307,7 +309,11 @@ def _dense_to_gid_from(curr: torch.Tensor, dense: torch.Tensor,
 
 
 def _modularity(state: PhaseState) -> float:
-    parts = modularity_parts(state.cluster_weight, state.local_degree)
+    if getattr(state, "use_hip", False):
+        from . import ops
+        parts = ops.modularity_parts(state.cluster_weight, state.local_degree)
+    else:
+        parts = modularity_parts(state.cluster_weight, state.local_degree)
     state.comm.allreduce_sum_(parts)
     c = state.constant
     return float(parts[0]) * c - float(parts[1]) * c * c

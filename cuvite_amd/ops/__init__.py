@@ -33,13 +33,58 @@ def _require():
     return _ext
 
 
+# Degree-class boundaries (see louvain_kernels.hip header comment).
+_CLASS_BOUNDS = (16, 64, 512, 4096)
+_bucket_cache: dict = {}
+
+
+def _buckets_for(rowptr: torch.Tensor):
+    """Degree-class vertex lists + hub-table offsets for a CSR (static per
+    phase; cached by the rowptr storage)."""
+    key = (rowptr.data_ptr(), rowptr.numel())
+    hit = _bucket_cache.get(key)
+    if hit is not None and hit[0] is rowptr:  # identity check: ptr reuse safe
+        return hit[1], hit[2]
+    deg = rowptr[1:] - rowptr[:-1]
+    b0, b1, b2, b3 = _CLASS_BOUNDS
+    vlists = [
+        ((deg > 0) & (deg <= b0)).nonzero(as_tuple=True)[0].to(torch.int32),
+        ((deg > b0) & (deg <= b1)).nonzero(as_tuple=True)[0].to(torch.int32),
+        ((deg > b1) & (deg <= b2)).nonzero(as_tuple=True)[0].to(torch.int32),
+        ((deg > b2) & (deg <= b3)).nonzero(as_tuple=True)[0].to(torch.int32),
+        (deg > b3).nonzero(as_tuple=True)[0].to(torch.int32),
+    ]
+    hubs = vlists[4]
+    if hubs.numel():
+        hdeg = deg[hubs.to(torch.int64)]
+        caps = torch.pow(
+            2.0, torch.ceil(torch.log2((2 * (hdeg + 1)).to(torch.float64)))
+        ).to(torch.int64)
+        offsets = torch.zeros(hubs.numel() + 1, dtype=torch.int64,
+                              device=rowptr.device)
+        offsets[1:] = torch.cumsum(caps, dim=0)
+    else:
+        offsets = torch.zeros(1, dtype=torch.int64, device=rowptr.device)
+    if len(_bucket_cache) > 8:
+        _bucket_cache.clear()
+    _bucket_cache[key] = (rowptr, vlists, offsets)
+    return vlists, offsets
+
+
 def local_move(inp):
     """HIP local-move iteration (see local_move.MoveInputs for semantics).
     Returns (target dense comm ids [nv], cluster_weight [nv])."""
     ext = _require()
-    return ext.local_move(
+    vlists, offsets = _buckets_for(inp.rowptr)
+    n_pool = int(offsets[-1])
+    dev = inp.rowptr.device
+    pool_keys = torch.full((n_pool,), -1, dtype=torch.int32, device=dev)
+    pool_vals = torch.zeros(n_pool, dtype=inp.weights.dtype, device=dev)
+    target, cw = ext.local_move_bucketed(
         inp.rowptr, inp.tails, inp.weights, inp.curr_comm, inp.v_degree,
-        inp.comm_size, inp.comm_degree, inp.comm_gid, float(inp.constant))
+        inp.comm_size, inp.comm_degree, inp.comm_gid, float(inp.constant),
+        vlists, offsets, pool_keys, pool_vals)
+    return target, cw
 
 
 def modularity_parts(cluster_weight: torch.Tensor,
