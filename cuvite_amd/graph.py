@@ -130,8 +130,9 @@ class Graph:
         lsrc1, dst1, w1 = lsrc[o1], dst[o1], w[o1]
         o2 = torch.argsort(lsrc1, stable=True)
         lsrc2, dst2, w2 = lsrc1[o2], dst1[o2], w1[o2]
-        rowptr = torch.zeros(nv + 1, dtype=torch.int64)
-        rowptr[1:] = torch.cumsum(torch.bincount(lsrc2, minlength=nv), dim=0)
+        rowptr = torch.zeros(nv + 1, dtype=torch.int64, device=src.device)
+        if lsrc2.numel():
+            rowptr[1:] = torch.cumsum(torch.bincount(lsrc2, minlength=nv), dim=0)
         return Graph(rowptr, dst2, w2)
 
 
