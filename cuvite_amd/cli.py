@@ -1,0 +1,273 @@
+"""Command-line driver: the MI355X-native equivalent of the reference's
+`graphClustering` binary (main.cpp:63-585, 587-712).
+
+Flag surface mirrors the reference's getopt string "f:bc:od:r:t:a:ig:zpn:e:s:j"
+with the same semantics, plus long-form extras for the MI355X build (R-MAT
+generation, device/backend selection). Distributed runs launch one process
+per GPU via `torch.distributed.run` (RCCL over xGMI on a GPU node, gloo on
+CPU); single-process runs need no launcher.
+
+Examples:
+    python -m cuvite_amd -f karate.bin -c 8 -i
+    python -m cuvite_amd -n 16384 -e 5 -o
+    torchrun --standalone --nproc-per-node 8 -m cuvite_amd --rmat 26
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+import time
+
+import torch
+
+from .compare import compare_communities
+from .generators import rgg_dist_graph, rmat_edges, rmat_edges_torch, karate_graph
+from .graph import DistGraph, Graph, Partition, single_partition
+from .io import (load_dist_graph, load_ground_truth, write_communities,
+                 write_dist_graph)
+from .louvain import LouvainConfig, louvain
+from .parallel import Comm, init_from_env
+from .utils.stats import print_dist_stats
+
+
+def build_parser() -> argparse.ArgumentParser:
+    ap = argparse.ArgumentParser(
+        prog="cuvite_amd",
+        description="Distributed Louvain community detection on MI355X "
+                    "(flag-compatible with the reference graphClustering; "
+                    "ref main.cpp:587-712)")
+    ap.add_argument("-f", dest="input", metavar="FILE", default="",
+                    help="input binary graph (Vite format)")
+    ap.add_argument("-b", dest="balanced", action="store_true",
+                    help="edge-balanced 1-D partition for file inputs")
+    ap.add_argument("-c", dest="coloring", metavar="NCOLORS", type=int,
+                    default=0, help="coloring-ordered moves, sync per color")
+    ap.add_argument("-d", dest="ordering", metavar="NCOLORS", type=int,
+                    default=0, help="color-ordered moves, no per-color sync")
+    ap.add_argument("-o", dest="output", action="store_true",
+                    help="write <input>.communities")
+    ap.add_argument("-r", dest="ranks_per_node", metavar="N", type=int,
+                    default=1, help="I/O aggregator hint (accepted; POSIX "
+                    "pread per rank needs no aggregation)")
+    ap.add_argument("-t", dest="early_term", metavar="TYPE", type=int,
+                    default=0, help="early termination type 1-4")
+    ap.add_argument("-a", dest="et_alpha", metavar="ALPHA", type=float,
+                    default=1.0, help="early termination alpha (types 2/4)")
+    ap.add_argument("-i", dest="threshold_cycling", action="store_true",
+                    help="threshold cycling across phases")
+    ap.add_argument("-g", dest="ground_truth", metavar="FILE", default="",
+                    help="ground-truth community file for comparison")
+    ap.add_argument("-z", dest="one_based", action="store_true",
+                    help="ground-truth file is 1-based")
+    ap.add_argument("-p", dest="one_phase", action="store_true",
+                    help="run a single Louvain phase")
+    ap.add_argument("-n", dest="gen_nv", metavar="NV", type=int, default=0,
+                    help="generate an in-memory RGG with NV vertices")
+    ap.add_argument("-e", dest="random_edge_percent", metavar="PCT",
+                    type=float, default=0.0,
+                    help="add PCT%% random edges to the generated graph")
+    ap.add_argument("-s", dest="gen_out", metavar="FILE", default="",
+                    help="write the generated graph to FILE (Vite binary)")
+    ap.add_argument("-j", dest="just_process", action="store_true",
+                    help="load/generate the graph, print stats, exit")
+    # MI355X-native extras
+    ap.add_argument("--rmat", metavar="SCALE", type=int, default=0,
+                    help="generate an R-MAT graph of 2^SCALE vertices")
+    ap.add_argument("--edgefactor", type=int, default=16)
+    ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--karate", action="store_true",
+                    help="use the built-in Zachary karate club graph")
+    ap.add_argument("--device", choices=["auto", "cpu", "cuda"],
+                    default="auto")
+    ap.add_argument("--backend", choices=["auto", "hip", "torch"],
+                    default="auto", help="local-move backend")
+    ap.add_argument("--threshold", type=float, default=1.0e-6)
+    ap.add_argument("--max-phases", type=int, default=0,
+                    help="cap the number of phases (0 = reference default)")
+    ap.add_argument("--stats", action="store_true",
+                    help="print the graph distribution table "
+                    "(ref printStats, distgraph.hpp:100-149)")
+    return ap
+
+
+def validate(args, world: int):
+    if args.coloring and args.ordering:
+        sys.exit("Cannot enable both -c and -d")
+    if args.one_phase and args.threshold_cycling:
+        sys.exit("Cannot enable both -p and -i")
+    if args.early_term and not 1 <= args.early_term <= 4:
+        sys.exit("-t must be 1..4")
+    if args.early_term in (2, 4) and not 0.0 <= args.et_alpha <= 1.0:
+        sys.exit("-a must be in [0,1]")
+    n_sources = sum(bool(x) for x in
+                    (args.input, args.gen_nv, args.rmat, args.karate))
+    if n_sources != 1:
+        sys.exit("Specify exactly one graph source: -f FILE, -n NV, "
+                 "--rmat SCALE, or --karate")
+    if args.random_edge_percent and not args.gen_nv:
+        sys.exit("-e needs -n (generated graph)")
+    if args.gen_nv and args.gen_nv % world != 0:
+        sys.exit("-n NV must be divisible by the process count")
+
+
+def _ingest(args, comm: Comm) -> DistGraph:
+    dev = comm.device
+    wdtype = torch.float64
+    if args.input:
+        return load_dist_graph(args.input, comm.rank, comm.world,
+                               balanced=args.balanced,
+                               weight_dtype=wdtype).to(dev)
+    if args.karate:
+        g = karate_graph(wdtype)
+        part = Partition.contiguous(g.nv, comm.world)
+        # re-slice the single CSR by partition for multi-rank runs
+        if comm.world == 1:
+            return DistGraph(g, part, comm.rank).to(dev)
+        base, bound = int(part.parts[comm.rank]), int(part.parts[comm.rank + 1])
+        e0, e1 = int(g.rowptr[base]), int(g.rowptr[bound])
+        lg = Graph(g.rowptr[base:bound + 1] - g.rowptr[base],
+                   g.tails[e0:e1], g.weights[e0:e1])
+        return DistGraph(lg, part, comm.rank).to(dev)
+    if args.gen_nv:
+        dg = rgg_dist_graph(args.gen_nv, comm.rank, comm.world,
+                            seed=args.seed,
+                            random_edge_percent=args.random_edge_percent,
+                            weight_dtype=wdtype)
+        if args.gen_out:
+            write_dist_graph_collect(args.gen_out, dg, comm)
+        return dg.to(dev)
+    # --rmat
+    nv = 1 << args.rmat
+    part = Partition.contiguous(nv, comm.world)
+    lo = comm.rank / comm.world
+    hi = (comm.rank + 1) / comm.world
+    if dev.type == "cuda":
+        u, v, w = rmat_edges_torch(args.rmat, args.edgefactor, args.seed,
+                                   lo, hi, dev, weight_dtype=wdtype)
+    else:
+        import numpy as np
+        uu, vv, ww = rmat_edges(args.rmat, args.edgefactor, args.seed, lo, hi)
+        u, v = torch.from_numpy(uu), torch.from_numpy(vv)
+        w = torch.from_numpy(ww).to(wdtype)
+    src = torch.cat([u, v])
+    dst = torch.cat([v, u])
+    ww = torch.cat([w, w])
+    if comm.world > 1:
+        parts_dev = part.parts.to(dev)
+        order = torch.argsort(src)
+        src, dst, ww = src[order], dst[order], ww[order]
+        offs = torch.searchsorted(src, parts_dev)
+        chunks = [src[offs[p]:offs[p + 1]] for p in range(comm.world)]
+        got_s = comm.all_to_all_v(chunks)
+        cnts = [int(t.numel()) for t in got_s]
+        got_d = comm.all_to_all_v(
+            [dst[offs[p]:offs[p + 1]] for p in range(comm.world)], cnts)
+        got_w = comm.all_to_all_v(
+            [ww[offs[p]:offs[p + 1]] for p in range(comm.world)], cnts)
+        src, dst, ww = torch.cat(got_s), torch.cat(got_d), torch.cat(got_w)
+    g = Graph.from_edge_tuples(part.nv_local(comm.rank), src, dst, ww,
+                               base=part.base(comm.rank))
+    return DistGraph(g, part, comm.rank)
+
+
+def write_dist_graph_collect(path: str, dg: DistGraph, comm: Comm):
+    """Root-gathers shards and writes one Vite binary (ref writeGraph,
+    distgraph.cpp:936-1014; cold path, host-side)."""
+    dev = comm.device
+    rp = comm.gather_cat(dg.g.rowptr.to(dev), root=0)
+    tl = comm.gather_cat(dg.g.tails.to(dev), root=0)
+    wt = comm.gather_cat(dg.g.weights.to(dev), root=0)
+    if comm.rank == 0:
+        shards = []
+        vo = 0
+        eo = 0
+        for p in range(comm.world):
+            nvp = dg.partition.nv_local(p)
+            rpp = rp[vo:vo + nvp + 1].cpu()
+            nep = int(rpp[-1])
+            shards.append(DistGraph(
+                Graph(rpp, tl[eo:eo + nep].cpu(), wt[eo:eo + nep].cpu()),
+                dg.part, p))
+            vo += nvp + 1
+            eo += nep
+        write_dist_graph(path, shards)
+        print(f"Wrote generated graph to {path}")
+
+
+def main(argv=None) -> int:
+    args = build_parser().parse_args(argv)
+    comm = init_from_env(prefer=args.device)
+    validate(args, comm.world)
+
+    t0 = time.perf_counter()
+    dg = _ingest(args, comm)
+    t_ingest = time.perf_counter() - t0
+    ne_global = comm.allreduce_scalar(float(dg.ne))
+    if comm.rank == 0:
+        print(f"Graph: nv={dg.nv_global} ne(directed)={int(ne_global)} "
+              f"ranks={comm.world} device={comm.device.type} "
+              f"ingest={t_ingest:.3f}s")
+    if args.stats:
+        print_dist_stats(dg, comm)
+    if args.just_process:
+        return 0
+
+    cfg = LouvainConfig(
+        threshold=args.threshold,
+        threshold_scaling=args.threshold_cycling,
+        one_phase=args.one_phase,
+        early_term=args.early_term,
+        et_delta=args.et_alpha,
+        coloring=args.coloring > 0,
+        ordering=args.ordering > 0,
+        max_colors=max(args.coloring, args.ordering) or 8,
+        backend=args.backend,
+    )
+    if args.max_phases:
+        cfg.max_phases = args.max_phases
+
+    comm.barrier()
+    t0 = time.perf_counter()
+    res = louvain(dg, comm, cfg)
+    comm.barrier()
+    t_total = time.perf_counter() - t0
+
+    if comm.rank == 0:
+        for lvl, q in enumerate(res.modularity_per_level):
+            print(f"Level {lvl}: modularity = {q:.6f}")
+        # TEPS per the reference definition (main.cpp:448,509):
+        # edges traversed = global directed edge count x total iterations
+        teps = ne_global * res.total_iters / t_total if t_total > 0 else 0.0
+        print(f"Final modularity: {res.modularity:.6f}")
+        print(f"Phases: {res.phases}  Iterations: {res.total_iters}")
+        print(f"Coloring time: {res.times.get('coloring', 0.0):.3f}s")
+        print(f"Clustering time: {res.times.get('clustering', 0.0):.3f}s")
+        print(f"Rebuild time: {res.times.get('rebuild', 0.0):.3f}s")
+        print(f"Total time: {t_total:.3f}s  TEPS: {teps:.4g}")
+
+    if args.output or args.ground_truth:
+        # gather final communities to root in vertex order
+        allc = comm.gather_cat(res.communities.to(comm.device), root=0)
+        if comm.rank == 0:
+            allc = allc.cpu()
+            if args.output:
+                out_path = (args.input or args.gen_out or "graph") + \
+                    ".communities"
+                write_communities(out_path, allc)
+                print(f"Wrote communities to {out_path}")
+            if args.ground_truth:
+                truth = load_ground_truth(args.ground_truth,
+                                          zero_based=not args.one_based)
+                m = compare_communities(truth, allc)
+                print(f"Ground truth: precision={m['precision']:.4f} "
+                      f"recall={m['recall']:.4f} f-score={m['f_score']:.4f} "
+                      f"f-mean={m['f_mean']:.4f} "
+                      f"gini(pred)={m['gini_pred']:.4f} "
+                      f"gini(truth)={m['gini_truth']:.4f}")
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

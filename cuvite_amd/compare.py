@@ -14,8 +14,9 @@ import numpy as np
 import torch
 
 
-def _best_f_mean(truth: np.ndarray, pred: np.ndarray) -> float:
-    """Mean over truth communities of the best F-score against pred."""
+def _best_f_mean(truth: np.ndarray, pred: np.ndarray):
+    """Mean over truth communities of the best F-score against pred, plus
+    the mean precision/recall of those best-F pairs."""
     # contingency via sparse counting on (truth, pred) pairs
     t_ids, t_inv = np.unique(truth, return_inverse=True)
     p_ids, p_inv = np.unique(pred, return_inverse=True)
@@ -31,7 +32,12 @@ def _best_f_mean(truth: np.ndarray, pred: np.ndarray) -> float:
     f = 2 * prec * rec / np.maximum(prec + rec, 1e-300)
     best = np.zeros(nt)
     np.maximum.at(best, ti, f)
-    return float(best.mean())
+    # precision/recall of the pair achieving the best F per truth community
+    order = np.argsort(f, kind="stable")
+    best_pair = np.zeros(nt, dtype=np.int64)
+    best_pair[ti[order]] = order  # last write per ti = argmax f
+    return float(best.mean()), float(prec[best_pair].mean()), \
+        float(rec[best_pair].mean())
 
 
 def compare_communities(truth: torch.Tensor, pred: torch.Tensor) -> dict:
@@ -40,12 +46,15 @@ def compare_communities(truth: torch.Tensor, pred: torch.Tensor) -> dict:
     t = truth.cpu().numpy()
     p = pred.cpu().numpy()
     assert t.shape == p.shape
-    f_tp = _best_f_mean(t, p)
-    f_pt = _best_f_mean(p, t)
+    f_tp, prec, rec = _best_f_mean(t, p)
+    f_pt, _, _ = _best_f_mean(p, t)
     return {
         "f_truth_to_pred": f_tp,
         "f_pred_to_truth": f_pt,
         "f_mean": 0.5 * (f_tp + f_pt),
+        "precision": prec,
+        "recall": rec,
+        "f_score": f_tp,
         "gini_truth": gini_coefficient(t),
         "gini_pred": gini_coefficient(p),
         "n_truth": int(len(np.unique(t))),

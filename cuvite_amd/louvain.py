@@ -214,7 +214,7 @@ def run_phase(dg: DistGraph, comm: Comm, cfg: LouvainConfig,
     while iters < cfg.max_iters_per_phase:
         iters += 1
         target = _one_sweep(state, cfg, move_fn,
-                            color_order if use_colors and cfg.coloring else None)
+                            color_order if use_colors else None)
 
         # modularity over the whole sweep
         curr_mod = _modularity(state)
@@ -260,8 +260,12 @@ def _one_sweep(state: PhaseState, cfg: LouvainConfig, move_fn,
         state.apply_moves(target, remote_gids)
         return target
 
-    # coloring-ordered: process color classes sequentially, adopting moves
-    # after each class (labels change within the sweep)
+    if cfg.ordering and not cfg.coloring:
+        return _ordered_sweep(state, cfg, move_fn, color_order)
+
+    # coloring-ordered (-c): process color classes sequentially, adopting
+    # moves after each class, with a fresh ghost-label exchange and remote
+    # community-info fetch per class (ref louvain.cpp:862-901).
     work = state.curr_comm.clone()
     cw_total = torch.zeros(nv, dtype=state.dg.g.weights.dtype, device=dev)
     for vidx in color_order:
@@ -290,6 +294,49 @@ def _one_sweep(state: PhaseState, cfg: LouvainConfig, move_fn,
         work = tgt
     state.cluster_weight = cw_total
     return work
+
+
+def _ordered_sweep(state: PhaseState, cfg: LouvainConfig, move_fn,
+                   color_order: List[torch.Tensor]):
+    """Vertex-ordering variant (-d, ref distLouvainMethodVertexOrder,
+    louvain.cpp:1433-1562): color classes are processed sequentially but
+    WITHOUT per-class communication — ghost labels and remote community
+    aggregates go stale within the sweep; one exchange at sweep start, one
+    delta push at sweep end. All per-class updates happen in the dense
+    community id space on device."""
+    nv = state.dg.nv
+    dev = state.dg.g.device
+    ghost_comm = exchange_ghost_labels(state.halo, state.curr_comm)
+    dense, remote_gids, c_size, c_degree, c_gid = state.densify(ghost_comm)
+    work_dense = dense.clone()
+    cw_total = torch.zeros(nv, dtype=state.dg.g.weights.dtype, device=dev)
+    for vidx in color_order:
+        inp = MoveInputs(state.dg.g.rowptr, state.halo.tails_dense,
+                         state.dg.g.weights, work_dense, state.v_degree,
+                         c_size, c_degree, c_gid, state.constant)
+        tgt_dense, cw = move_fn(inp)
+        mask = torch.zeros(nv, dtype=torch.bool, device=dev)
+        mask[vidx] = True
+        if cfg.early_term:
+            mask &= state.active
+        new_local = torch.where(mask, tgt_dense, work_dense[:nv])
+        cw_total = torch.where(mask, cw, cw_total)
+        # local-view aggregate update in dense space (no communication)
+        moved = new_local != work_dense[:nv]
+        if bool(moved.any()):
+            src = work_dense[:nv][moved].to(torch.int64)
+            dst = new_local[moved].to(torch.int64)
+            vdeg = state.v_degree[moved]
+            ones = torch.ones_like(src)
+            c_size.index_add_(0, src, -ones)
+            c_size.index_add_(0, dst, ones)
+            c_degree.index_add_(0, src, -vdeg)
+            c_degree.index_add_(0, dst, vdeg)
+        work_dense[:nv] = new_local
+    state.cluster_weight = cw_total
+    target = _dense_to_gid(state, work_dense[:nv], remote_gids)
+    state.apply_moves(target, remote_gids)
+    return target
 
 
 def _dense_to_gid(state: PhaseState, dense: torch.Tensor,
@@ -356,7 +403,7 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
         t0 = time.perf_counter()
         curr_mod, cvect, iters = run_phase(
             level, comm, cfg, curr_mod, threshold,
-            colors=colors if cfg.coloring else None, num_colors=num_colors)
+            colors=colors, num_colors=num_colors)
         times["clustering"] += time.perf_counter() - t0
         tot_iters += iters
 

@@ -144,20 +144,22 @@ class Comm:
 
 
 def init_from_env(backend: Optional[str] = None,
-                  timeout_s: int = 600) -> Comm:
+                  timeout_s: int = 600, prefer: str = "auto") -> Comm:
     """Initialize torch.distributed from torchrun env vars and pick the GPU
     for this rank (ref set_gpuDevices, louvain_cuda.cu:1634-1669: one device
-    per node-local rank)."""
+    per node-local rank). prefer: "auto" (GPU if present), "cpu", "cuda"."""
+    use_cuda = torch.cuda.is_available() if prefer == "auto" \
+        else prefer == "cuda"
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if world == 1:
-        dev = torch.device("cuda:0") if torch.cuda.is_available() else torch.device("cpu")
+        dev = torch.device("cuda:0") if use_cuda else torch.device("cpu")
         if dev.type == "cuda":
             torch.cuda.set_device(dev)
         return Comm(dev)
     rank = int(os.environ["RANK"])
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = "nccl" if use_cuda else "gloo"
     if backend == "nccl":
         torch.cuda.set_device(local_rank % torch.cuda.device_count())
         device = torch.device("cuda", local_rank % torch.cuda.device_count())
