@@ -26,53 +26,10 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-from cuvite_amd.generators import rmat_edges_torch, rmat_edges  # noqa: E402
-from cuvite_amd.graph import Graph, DistGraph, Partition  # noqa: E402
+from cuvite_amd.generators import rmat_dist_graph  # noqa: E402
 from cuvite_amd.louvain import (LouvainConfig, PhaseState, _modularity,
                                 _one_sweep, _pick_move_fn)  # noqa: E402
 from cuvite_amd.parallel import Comm, init_from_env  # noqa: E402
-
-
-def build_rmat_dist(scale, edgefactor, seed, comm, device, wdtype):
-    """Generate this rank's R-MAT slice on device and route directed edges to
-    their 1-D owners."""
-    nv = 1 << scale
-    part = Partition.contiguous(nv, comm.world)
-    lo = comm.rank / comm.world
-    hi = (comm.rank + 1) / comm.world
-    if device.type == "cuda":
-        u, v, w = rmat_edges_torch(scale, edgefactor, seed, lo, hi, device,
-                                   weight_dtype=wdtype)
-    else:
-        import numpy as np
-        uu, vv, ww = rmat_edges(scale, edgefactor, seed, lo, hi)
-        u = torch.from_numpy(uu)
-        v = torch.from_numpy(vv)
-        w = torch.from_numpy(ww).to(wdtype)
-    # symmetrize: directed copies
-    src = torch.cat([u, v])
-    dst = torch.cat([v, u])
-    ww = torch.cat([w, w])
-    del u, v, w
-    if comm.world > 1:
-        parts_dev = part.parts.to(device)
-        order = torch.argsort(src)
-        src, dst, ww = src[order], dst[order], ww[order]
-        del order
-        offs = torch.searchsorted(src, parts_dev)
-        sp_s = [src[offs[p]:offs[p + 1]] for p in range(comm.world)]
-        got_s = comm.all_to_all_v(sp_s)
-        cnts = [int(g.numel()) for g in got_s]
-        got_d = comm.all_to_all_v(
-            [dst[offs[p]:offs[p + 1]] for p in range(comm.world)], cnts)
-        got_w = comm.all_to_all_v(
-            [ww[offs[p]:offs[p + 1]] for p in range(comm.world)], cnts)
-        src = torch.cat(got_s)
-        dst = torch.cat(got_d)
-        ww = torch.cat(got_w)
-    g = Graph.from_edge_tuples(part.nv_local(comm.rank), src, dst, ww,
-                               base=part.base(comm.rank))
-    return DistGraph(g, part, comm.rank)
 
 
 def main():
@@ -93,8 +50,8 @@ def main():
     wdtype = torch.float64 if args.dtype == "fp64" else torch.float32
 
     t_gen0 = time.perf_counter()
-    dg = build_rmat_dist(args.scale, args.edgefactor, args.seed, comm, device,
-                         wdtype)
+    dg = rmat_dist_graph(args.scale, args.edgefactor, args.seed, comm,
+                         device, weight_dtype=wdtype)
     ne_global = float(comm.allreduce_scalar(float(dg.ne)))
     t_gen = time.perf_counter() - t_gen0
 

@@ -23,7 +23,7 @@ import time
 import torch
 
 from .compare import compare_communities
-from .generators import rgg_dist_graph, rmat_edges, rmat_edges_torch, karate_graph
+from .generators import rgg_dist_graph, rmat_dist_graph, karate_graph
 from .graph import DistGraph, Graph, Partition, single_partition
 from .io import (load_dist_graph, load_ground_truth, write_communities,
                  write_dist_graph)
@@ -139,37 +139,8 @@ def _ingest(args, comm: Comm) -> DistGraph:
             write_dist_graph_collect(args.gen_out, dg, comm)
         return dg.to(dev)
     # --rmat
-    nv = 1 << args.rmat
-    part = Partition.contiguous(nv, comm.world)
-    lo = comm.rank / comm.world
-    hi = (comm.rank + 1) / comm.world
-    if dev.type == "cuda":
-        u, v, w = rmat_edges_torch(args.rmat, args.edgefactor, args.seed,
-                                   lo, hi, dev, weight_dtype=wdtype)
-    else:
-        import numpy as np
-        uu, vv, ww = rmat_edges(args.rmat, args.edgefactor, args.seed, lo, hi)
-        u, v = torch.from_numpy(uu), torch.from_numpy(vv)
-        w = torch.from_numpy(ww).to(wdtype)
-    src = torch.cat([u, v])
-    dst = torch.cat([v, u])
-    ww = torch.cat([w, w])
-    if comm.world > 1:
-        parts_dev = part.parts.to(dev)
-        order = torch.argsort(src)
-        src, dst, ww = src[order], dst[order], ww[order]
-        offs = torch.searchsorted(src, parts_dev)
-        chunks = [src[offs[p]:offs[p + 1]] for p in range(comm.world)]
-        got_s = comm.all_to_all_v(chunks)
-        cnts = [int(t.numel()) for t in got_s]
-        got_d = comm.all_to_all_v(
-            [dst[offs[p]:offs[p + 1]] for p in range(comm.world)], cnts)
-        got_w = comm.all_to_all_v(
-            [ww[offs[p]:offs[p + 1]] for p in range(comm.world)], cnts)
-        src, dst, ww = torch.cat(got_s), torch.cat(got_d), torch.cat(got_w)
-    g = Graph.from_edge_tuples(part.nv_local(comm.rank), src, dst, ww,
-                               base=part.base(comm.rank))
-    return DistGraph(g, part, comm.rank)
+    return rmat_dist_graph(args.rmat, args.edgefactor, args.seed, comm, dev,
+                           weight_dtype=wdtype)
 
 
 def write_dist_graph_collect(path: str, dg: DistGraph, comm: Comm):
@@ -189,7 +160,7 @@ def write_dist_graph_collect(path: str, dg: DistGraph, comm: Comm):
             nep = int(rpp[-1])
             shards.append(DistGraph(
                 Graph(rpp, tl[eo:eo + nep].cpu(), wt[eo:eo + nep].cpu()),
-                dg.part, p))
+                dg.partition, p))
             vo += nvp + 1
             eo += nep
         write_dist_graph(path, shards)

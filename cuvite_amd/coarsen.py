@@ -137,10 +137,14 @@ def _aggregate(s: torch.Tensor, t: torch.Tensor, w: torch.Tensor, gnc: int):
     key = s * gnc + t
     key_s, order = torch.sort(key)
     w_s = w[order]
-    uniq, inv = torch.unique_consecutive(key_s, return_inverse=True)
-    w_out = torch.zeros(uniq.numel(), dtype=w.dtype, device=w.device)
-    w_out.index_add_(0, inv, w_s)
-    return uniq // gnc, uniq % gnc, w_out
+    uniq, counts = torch.unique_consecutive(key_s, return_counts=True)
+    # segmented sum over sorted runs via cumsum (no atomics: torch fp64
+    # scatter-adds are CAS loops on ROCm, and this is deterministic)
+    ends = torch.cumsum(counts, dim=0) - 1
+    cs = torch.cumsum(w_s.to(torch.float64), dim=0)
+    w_out = cs[ends].clone()
+    w_out[1:] -= cs[ends[:-1]]
+    return uniq // gnc, uniq % gnc, w_out.to(w.dtype)
 
 
 def remap_labels(dg: DistGraph, comm: Comm, assign: torch.Tensor,

@@ -156,6 +156,79 @@ def rmat_edges_torch(scale: int, edgefactor: int, seed: int,
     return torch.cat(us), torch.cat(vs), torch.cat(ws)
 
 
+_ROUTE_CHUNK = 1 << 28  # 256M tuples per routing slice (torch op INT_MAX cap)
+
+
+def rmat_dist_graph(scale: int, edgefactor: int, seed: int, comm,
+                    device, weight_dtype: torch.dtype = torch.float64):
+    """This rank's shard of the symmetrized R-MAT graph: generate the rank's
+    slice of the global undirected edge list, emit both directed copies, and
+    route each to its 1-D owner. Routing is chunked owner-bucketing (no
+    global argsort: torch.sort is capped at INT_MAX elements, and a scale-26
+    symmetrized list is 2.1e9 tuples)."""
+    from .graph import DistGraph, Graph, Partition
+
+    nv = 1 << scale
+    part = Partition.contiguous(nv, comm.world)
+    lo = comm.rank / comm.world
+    hi = (comm.rank + 1) / comm.world
+    if device.type == "cuda":
+        u, v, w = rmat_edges_torch(scale, edgefactor, seed, lo, hi, device,
+                                   weight_dtype=weight_dtype)
+    else:
+        uu, vv, ww = rmat_edges(scale, edgefactor, seed, lo, hi)
+        u, v = torch.from_numpy(uu), torch.from_numpy(vv)
+        w = torch.from_numpy(ww).to(weight_dtype)
+
+    if comm.world == 1:
+        src = torch.cat([u, v])
+        dst = torch.cat([v, u])
+        ww = torch.cat([w, w])
+        del u, v, w
+        g = Graph.from_edge_tuples(nv, src, dst, ww)
+        return DistGraph(g, part, comm.rank)
+
+    parts_dev = part.parts.to(device)
+    kept_s, kept_d, kept_w = [], [], []
+    n = u.numel()
+    # every rank must run the same number of collective rounds
+    n_max = int(comm.allreduce_scalar(float(n), op="max"))
+    for c0 in range(0, max(n_max, 1), _ROUTE_CHUNK):
+        c1 = min(c0 + _ROUTE_CHUNK, n)
+        c0 = min(c0, c1)
+        # both directed copies of this chunk
+        cs = torch.cat([u[c0:c1], v[c0:c1]])
+        cd = torch.cat([v[c0:c1], u[c0:c1]])
+        cw = torch.cat([w[c0:c1], w[c0:c1]])
+        owner = torch.searchsorted(parts_dev[1:], cs, right=True)
+        send_s, send_d, send_w = [], [], []
+        for p in range(comm.world):
+            m = owner == p
+            send_s.append(cs[m])
+            send_d.append(cd[m])
+            send_w.append(cw[m])
+        got_s = comm.all_to_all_v(send_s)
+        cnts = [int(t.numel()) for t in got_s]
+        got_d = comm.all_to_all_v(send_d, cnts)
+        got_w = comm.all_to_all_v(send_w, cnts)
+        kept_s.append(torch.cat(got_s))
+        kept_d.append(torch.cat(got_d))
+        kept_w.append(torch.cat(got_w))
+    del u, v, w
+    src = torch.cat(kept_s) if kept_s else torch.zeros(0, dtype=torch.int64,
+                                                       device=device)
+    del kept_s
+    dst = torch.cat(kept_d) if kept_d else torch.zeros(0, dtype=torch.int64,
+                                                       device=device)
+    del kept_d
+    ww = torch.cat(kept_w) if kept_w else torch.zeros(0, dtype=weight_dtype,
+                                                      device=device)
+    del kept_w
+    g = Graph.from_edge_tuples(part.nv_local(comm.rank), src, dst, ww,
+                               base=part.base(comm.rank))
+    return DistGraph(g, part, comm.rank)
+
+
 def rmat_graph(scale: int, edgefactor: int = 16, seed: int = 1,
                weight_dtype: torch.dtype = torch.float64) -> Graph:
     """Whole symmetrized R-MAT graph on one process (tests / small runs)."""

@@ -122,8 +122,16 @@ class Graph:
     def from_edge_tuples(nv: int, src: torch.Tensor, dst: torch.Tensor,
                          w: torch.Tensor, base: int = 0) -> "Graph":
         """Assemble a CSR from (src, dst, w) directed edge tuples where src are
-        global ids in [base, base+nv). Sorts by (src, dst) for determinism.
-        Reference analog: processGraphData (utils.cpp:10-87)."""
+        global ids in [base, base+nv). Reference analog: processGraphData
+        (utils.cpp:10-87). On GPU this uses the sort-free HIP builder
+        (histogram + atomic-cursor placement; torch.argsort is capped at
+        INT_MAX elements and a lexicographic sort is not needed — all CSR
+        consumers are row-order-invariant). The CPU path sorts by (src, dst)
+        for determinism."""
+        if src.is_cuda:
+            from . import ops
+            rowptr, tails, weights = ops.csr_from_edges(nv, base, src, dst, w)
+            return Graph(rowptr, tails, weights)
         lsrc = src - base
         # stable two-pass lexicographic sort: by dst then by src
         o1 = torch.argsort(dst, stable=True)
@@ -183,6 +191,9 @@ class DistGraph:
         nv = self.nv
         if self.g.ne == 0:
             return torch.zeros(nv, dtype=self.g.weights.dtype, device=self.g.device)
+        if self.g.device.type == "cuda":
+            from . import ops
+            return ops.row_sum(self.g.rowptr, self.g.weights)
         seg = torch.repeat_interleave(
             torch.arange(nv, device=self.g.device), self.g.degrees()
         )

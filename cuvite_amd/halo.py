@@ -22,6 +22,7 @@ from typing import List, Optional
 import torch
 
 from .graph import DistGraph
+from .ops import scatter_add_
 from .parallel import Comm
 
 
@@ -171,4 +172,4 @@ def push_remote_deltas(ctx: HaloContext, gids: torch.Tensor,
             continue
         li = got_ids[p] - dg.base
         local_size.index_add_(0, li, got_ds[p])
-        local_degree.index_add_(0, li, got_dd[p])
+        scatter_add_(local_degree, li, got_dd[p])

@@ -26,6 +26,7 @@ from .graph import DistGraph, Partition
 from .halo import (HaloContext, build_halo, exchange_ghost_labels,
                    fetch_remote_comm_info, push_remote_deltas)
 from .local_move import MoveInputs, local_move_torch, modularity_parts
+from .ops import scatter_add_
 from .parallel import Comm
 
 TERMINATION_PHASE_COUNT = 200   # ref utils.hpp:17-19
@@ -157,7 +158,7 @@ class PhaseState:
         is_local = (gids >= base) & (gids < bound)
         li = gids[is_local] - base
         self.local_size.index_add_(0, li, dsize[is_local])
-        self.local_degree.index_add_(0, li, ddeg[is_local])
+        scatter_add_(self.local_degree, li, ddeg[is_local])
         if self.comm.world > 1:
             push_remote_deltas(self.halo, gids[~is_local], dsize[~is_local],
                                ddeg[~is_local], self.local_size, self.local_degree)
@@ -330,8 +331,8 @@ def _ordered_sweep(state: PhaseState, cfg: LouvainConfig, move_fn,
             ones = torch.ones_like(src)
             c_size.index_add_(0, src, -ones)
             c_size.index_add_(0, dst, ones)
-            c_degree.index_add_(0, src, -vdeg)
-            c_degree.index_add_(0, dst, vdeg)
+            scatter_add_(c_degree, src, -vdeg)
+            scatter_add_(c_degree, dst, vdeg)
         work_dense[:nv] = new_local
     state.cluster_weight = cw_total
     target = _dense_to_gid(state, work_dense[:nv], remote_gids)

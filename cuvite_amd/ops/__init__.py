@@ -92,3 +92,31 @@ def modularity_parts(cluster_weight: torch.Tensor,
     """HIP reduction of (sum cluster_weight, sum degree^2) in fp64."""
     ext = _require()
     return ext.modularity_parts(cluster_weight, local_comm_degree)
+
+
+def scatter_add_(out: torch.Tensor, idx: torch.Tensor,
+                 val: torch.Tensor) -> torch.Tensor:
+    """out[idx[i]] += val[i]. On GPU float tensors this MUST go through the
+    HIP kernel (native fp64 atomics): torch's index_add_ lowers fp64 atomic
+    adds to a CAS loop on ROCm, measured 16.5 s per call at R-MAT s22
+    (profiles/bench_s22 first profile). Integer dtypes and CPU tensors use
+    torch's fast path."""
+    if out.is_cuda and out.is_floating_point():
+        _require().scatter_add_(out, idx, val.to(out.dtype))
+    else:
+        out.index_add_(0, idx, val.to(out.dtype))
+    return out
+
+
+def csr_from_edges(nv: int, base: int, src: torch.Tensor, dst: torch.Tensor,
+                   w: torch.Tensor):
+    """Sort-free device CSR assembly: (rowptr, tails, weights). Handles edge
+    counts beyond INT_MAX (torch.argsort cannot)."""
+    ext = _require()
+    return ext.csr_from_edges(nv, base, src, dst, w)
+
+
+def row_sum(rowptr: torch.Tensor, weights: torch.Tensor) -> torch.Tensor:
+    """Per-vertex weighted degree on device (one wave per CSR row)."""
+    ext = _require()
+    return ext.row_sum(rowptr, weights)

@@ -33,6 +33,16 @@ void launch_global(const int32_t*, int, const int64_t*, int32_t*, W*,
                    const MoveArgs<W>&, hipStream_t);
 template <typename W>
 void launch_modularity(const W*, const W*, int64_t, double*, hipStream_t);
+template <typename W>
+void launch_scatter_add(W*, const int64_t*, const W*, int64_t, hipStream_t);
+void launch_degree_count(const int64_t*, int64_t, int64_t, int32_t*,
+                         hipStream_t);
+template <typename W>
+void launch_csr_place(const int64_t*, const int64_t*, const W*, int64_t,
+                      int64_t, const int64_t*, int32_t*, int64_t*, W*,
+                      hipStream_t);
+template <typename W>
+void launch_row_sum(const int64_t*, const W*, int64_t, W*, hipStream_t);
 
 }  // namespace cuvite
 
@@ -125,6 +135,67 @@ at::Tensor modularity_parts(at::Tensor cluster_weight, at::Tensor comm_degree) {
   return out;
 }
 
+void scatter_add_(at::Tensor out, at::Tensor idx, at::Tensor val) {
+  CHECK_DEV(out); CHECK_CONT(out);
+  CHECK_DEV(idx); CHECK_CONT(idx);
+  CHECK_DEV(val); CHECK_CONT(val);
+  TORCH_CHECK(idx.scalar_type() == at::kLong, "idx must be int64");
+  TORCH_CHECK(idx.numel() == val.numel(), "idx/val length mismatch");
+  TORCH_CHECK(out.scalar_type() == val.scalar_type(), "dtype mismatch");
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES(out.scalar_type(), "scatter_add", [&] {
+    cuvite::launch_scatter_add<scalar_t>(out.data_ptr<scalar_t>(),
+                                         idx.data_ptr<int64_t>(),
+                                         val.data_ptr<scalar_t>(),
+                                         idx.numel(), stream);
+  });
+  C10_HIP_CHECK(hipGetLastError());
+}
+
+std::vector<at::Tensor> csr_from_edges(int64_t nv, int64_t base,
+                                       at::Tensor src, at::Tensor dst,
+                                       at::Tensor w) {
+  CHECK_DEV(src); CHECK_CONT(src);
+  CHECK_DEV(dst); CHECK_CONT(dst);
+  CHECK_DEV(w); CHECK_CONT(w);
+  TORCH_CHECK(src.scalar_type() == at::kLong && dst.scalar_type() == at::kLong,
+              "src/dst must be int64");
+  const int64_t ne = src.numel();
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto cnt = at::zeros({nv}, src.options().dtype(at::kInt));
+  cuvite::launch_degree_count(src.data_ptr<int64_t>(), ne, base,
+                              cnt.data_ptr<int32_t>(), stream);
+  auto rowptr = at::zeros({nv + 1}, src.options());
+  rowptr.narrow(0, 1, nv).copy_(at::cumsum(cnt, 0));
+  auto tails = at::empty({ne}, src.options());
+  auto weights = at::empty({ne}, w.options());
+  cnt.zero_();  // reuse as the per-row placement cursor
+  AT_DISPATCH_FLOATING_TYPES(w.scalar_type(), "csr_place", [&] {
+    cuvite::launch_csr_place<scalar_t>(
+        src.data_ptr<int64_t>(), dst.data_ptr<int64_t>(),
+        w.data_ptr<scalar_t>(), ne, base, rowptr.data_ptr<int64_t>(),
+        cnt.data_ptr<int32_t>(), tails.data_ptr<int64_t>(),
+        weights.data_ptr<scalar_t>(), stream);
+  });
+  C10_HIP_CHECK(hipGetLastError());
+  return {rowptr, tails, weights};
+}
+
+at::Tensor row_sum(at::Tensor rowptr, at::Tensor weights) {
+  CHECK_DEV(rowptr); CHECK_CONT(rowptr);
+  CHECK_DEV(weights); CHECK_CONT(weights);
+  const int64_t nv = rowptr.numel() - 1;
+  auto out = at::empty({nv}, weights.options());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES(weights.scalar_type(), "row_sum", [&] {
+    cuvite::launch_row_sum<scalar_t>(rowptr.data_ptr<int64_t>(),
+                                     weights.data_ptr<scalar_t>(), nv,
+                                     out.data_ptr<scalar_t>(), stream);
+  });
+  C10_HIP_CHECK(hipGetLastError());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -132,4 +203,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Louvain local-move iteration (HIP, degree-class routed)");
   m.def("modularity_parts", &modularity_parts,
         "fp64 (sum cw, sum degree^2) reduction (HIP)");
+  m.def("scatter_add_", &scatter_add_,
+        "out[idx[i]] += val[i] with native fp atomics (HIP)");
+  m.def("csr_from_edges", &csr_from_edges,
+        "sort-free CSR assembly on device (HIP)");
+  m.def("row_sum", &row_sum, "per-row CSR weight sum (HIP)");
 }

@@ -376,6 +376,76 @@ __global__ __launch_bounds__(BLOCK) void modularity_reduce(
   }
 }
 
+// ---------------------------------------------------------------------------
+// scatter_add: out[idx[i]] += val[i] with NATIVE fp32/fp64 atomics
+// (unsafeAtomicAdd -> global_atomic_add_f64 on gfx950). PyTorch-ROCm's
+// index_add_ lowers fp64 atomics to a CAS loop, which measured 16.5 s per
+// call on an R-MAT s22 sweep (profiles/); this kernel replaces it on the
+// hot paths (degree sums, community-aggregate deltas).
+// ---------------------------------------------------------------------------
+
+template <typename W>
+__global__ void scatter_add_kernel(W* __restrict__ out,
+                                   const int64_t* __restrict__ idx,
+                                   const W* __restrict__ val, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    unsafeAtomicAdd(&out[idx[i]], val[i]);
+}
+
+// ---------------------------------------------------------------------------
+// Device-side CSR assembly (no sort; replaces torch.argsort which is capped
+// at INT_MAX elements): degree histogram + atomic-cursor placement.
+// Reference analog: processGraphData (utils.cpp:10-87) done host-side there.
+// Row-internal edge order is nondeterministic; all consumers (local-move
+// hash tables, degree sums) are order-invariant.
+// ---------------------------------------------------------------------------
+
+__global__ void degree_count_kernel(const int64_t* __restrict__ src, int64_t n,
+                                    int64_t base, int32_t* __restrict__ cnt) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    atomicAdd(&cnt[src[i] - base], 1);
+}
+
+template <typename W>
+__global__ void csr_place_kernel(const int64_t* __restrict__ src,
+                                 const int64_t* __restrict__ dst,
+                                 const W* __restrict__ w, int64_t n,
+                                 int64_t base,
+                                 const int64_t* __restrict__ rowptr,
+                                 int32_t* __restrict__ cursor,
+                                 int64_t* __restrict__ tails_out,
+                                 W* __restrict__ w_out) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    const int64_t s = src[i] - base;
+    const int64_t o = rowptr[s] + atomicAdd(&cursor[s], 1);
+    tails_out[o] = dst[i];
+    w_out[o] = w[i];
+  }
+}
+
+// Per-vertex weighted degree: one wave per vertex, lane-strided row scan
+// (ref distSumVertexDegree, louvain.cpp:2126-2151).
+template <typename W, int BLOCK>
+__global__ __launch_bounds__(BLOCK) void row_sum_kernel(
+    const int64_t* __restrict__ rowptr, const W* __restrict__ weights,
+    int64_t nv, W* __restrict__ out) {
+  constexpr int WAVES = BLOCK / 64;
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const int64_t v = (int64_t)blockIdx.x * WAVES + wave;
+  if (v >= nv) return;
+  const int64_t e0 = rowptr[v], e1 = rowptr[v + 1];
+  double acc = 0.0;
+  for (int64_t e = e0 + lane; e < e1; e += 64) acc += (double)weights[e];
+  acc = sum_reduce<64>(acc);
+  if (lane == 0) out[v] = (W)acc;
+}
+
 // ------------------------------- launchers ---------------------------------
 
 template <typename W>
@@ -472,5 +542,64 @@ template void launch_modularity<float>(const float*, const float*, int64_t,
                                        double*, hipStream_t);
 template void launch_modularity<double>(const double*, const double*, int64_t,
                                         double*, hipStream_t);
+
+static int grid_for(int64_t n, int block) {
+  int64_t g = (n + block - 1) / block;
+  // >> 2048 workgroups keeps all 8 XCDs fed; cap to bound tail latency
+  return (int)(g < 1 ? 1 : (g > 65535 ? 65535 : g));
+}
+
+template <typename W>
+void launch_scatter_add(W* out, const int64_t* idx, const W* val, int64_t n,
+                        hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL((scatter_add_kernel<W>), dim3(grid_for(n, 256)),
+                     dim3(256), 0, stream, out, idx, val, n);
+}
+template void launch_scatter_add<float>(float*, const int64_t*, const float*,
+                                        int64_t, hipStream_t);
+template void launch_scatter_add<double>(double*, const int64_t*,
+                                         const double*, int64_t, hipStream_t);
+
+void launch_degree_count(const int64_t* src, int64_t n, int64_t base,
+                         int32_t* cnt, hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL(degree_count_kernel, dim3(grid_for(n, 256)), dim3(256), 0,
+                     stream, src, n, base, cnt);
+}
+
+template <typename W>
+void launch_csr_place(const int64_t* src, const int64_t* dst, const W* w,
+                      int64_t n, int64_t base, const int64_t* rowptr,
+                      int32_t* cursor, int64_t* tails_out, W* w_out,
+                      hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL((csr_place_kernel<W>), dim3(grid_for(n, 256)), dim3(256),
+                     0, stream, src, dst, w, n, base, rowptr, cursor,
+                     tails_out, w_out);
+}
+template void launch_csr_place<float>(const int64_t*, const int64_t*,
+                                      const float*, int64_t, int64_t,
+                                      const int64_t*, int32_t*, int64_t*,
+                                      float*, hipStream_t);
+template void launch_csr_place<double>(const int64_t*, const int64_t*,
+                                       const double*, int64_t, int64_t,
+                                       const int64_t*, int32_t*, int64_t*,
+                                       double*, hipStream_t);
+
+template <typename W>
+void launch_row_sum(const int64_t* rowptr, const W* weights, int64_t nv,
+                    W* out, hipStream_t stream) {
+  if (nv == 0) return;
+  constexpr int BLOCK = 256;
+  constexpr int WAVES = BLOCK / 64;
+  const int64_t grid = (nv + WAVES - 1) / WAVES;
+  hipLaunchKernelGGL((row_sum_kernel<W, BLOCK>), dim3((uint32_t)grid),
+                     dim3(BLOCK), 0, stream, rowptr, weights, nv, out);
+}
+template void launch_row_sum<float>(const int64_t*, const float*, int64_t,
+                                    float*, hipStream_t);
+template void launch_row_sum<double>(const int64_t*, const double*, int64_t,
+                                     double*, hipStream_t);
 
 }  // namespace cuvite

@@ -37,11 +37,12 @@ class Comm:
             dist.all_reduce(t, op=dist.ReduceOp.SUM)
         return t
 
-    def allreduce_scalar(self, x: float) -> float:
+    def allreduce_scalar(self, x: float, op: str = "sum") -> float:
         if not self.active or self.world == 1:
             return x
         t = torch.tensor([x], dtype=torch.float64, device=self.device)
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX if op == "max"
+                        else dist.ReduceOp.SUM)
         return float(t[0])
 
     def allgather_counts(self, counts: torch.Tensor) -> torch.Tensor:

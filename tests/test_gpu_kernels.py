@@ -6,7 +6,7 @@ import pytest
 import torch
 
 from cuvite_amd.generators import karate_graph, rmat_graph
-from cuvite_amd.graph import single_partition
+from cuvite_amd.graph import Graph, single_partition
 from cuvite_amd.local_move import (MoveInputs, local_move_torch,
                                    modularity_parts)
 
@@ -162,3 +162,50 @@ def test_native_extension_is_loaded():
     assert ops.available()
     import cuvite_amd.ops._hip_ops as ext
     assert "cuvite_amd/ops" in ext.__file__
+
+
+def test_scatter_add_fp64_matches_index_add():
+    from cuvite_amd import ops
+    dev = torch.device("cuda:0")
+    for dtype in (torch.float64, torch.float32):
+        out = torch.zeros(1000, dtype=dtype, device=dev)
+        ref = out.clone()
+        idx = torch.randint(0, 1000, (1 << 20,), device=dev)
+        val = torch.rand(1 << 20, dtype=dtype, device=dev)
+        ops.scatter_add_(out, idx, val)
+        ref.index_add_(0, idx, val)
+        tol = 1e-9 if dtype == torch.float64 else 1e-2
+        assert torch.allclose(out, ref, rtol=tol, atol=tol)
+
+
+def test_csr_from_edges_matches_cpu_build():
+    from cuvite_amd import ops
+    dev = torch.device("cuda:0")
+    torch.manual_seed(7)
+    nv, ne, base = 500, 20000, 100
+    src = torch.randint(base, base + nv, (ne,), device=dev)
+    dst = torch.randint(0, 5000, (ne,), device=dev)
+    w = torch.rand(ne, dtype=torch.float64, device=dev)
+    g_gpu = Graph.from_edge_tuples(nv, src, dst, w, base=base)
+    g_cpu = Graph.from_edge_tuples(nv, src.cpu(), dst.cpu(), w.cpu(), base=base)
+    assert torch.equal(g_gpu.rowptr.cpu(), g_cpu.rowptr)
+    # same multiset of (tail, weight) per row
+    for v in range(0, nv, 37):
+        e0, e1 = int(g_cpu.rowptr[v]), int(g_cpu.rowptr[v + 1])
+        a = sorted(zip(g_gpu.tails[e0:e1].cpu().tolist(),
+                       g_gpu.weights[e0:e1].cpu().tolist()))
+        b = sorted(zip(g_cpu.tails[e0:e1].tolist(),
+                       g_cpu.weights[e0:e1].tolist()))
+        assert a == b
+
+
+def test_row_sum_matches_degree():
+    from cuvite_amd import ops
+    dev = torch.device("cuda:0")
+    g = _unit_rmat(10).to(dev)
+    out = ops.row_sum(g.rowptr, g.weights)
+    seg = torch.repeat_interleave(
+        torch.arange(g.nv, device=dev), g.degrees())
+    ref = torch.zeros(g.nv, dtype=g.weights.dtype, device=dev)
+    ref.index_add_(0, seg, g.weights)
+    assert torch.allclose(out, ref, rtol=1e-12, atol=1e-12)
