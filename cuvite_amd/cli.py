@@ -23,7 +23,8 @@ import time
 import torch
 
 from .compare import compare_communities
-from .generators import rgg_dist_graph, rmat_dist_graph, karate_graph
+from .generators import (rgg_dist_graph, rmat_dist_graph, karate_graph,
+                         lfr_dist_graph)
 from .graph import DistGraph, Graph, Partition, single_partition
 from .io import (load_dist_graph, load_ground_truth, write_communities,
                  write_dist_graph)
@@ -79,6 +80,11 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--seed", type=int, default=1)
     ap.add_argument("--karate", action="store_true",
                     help="use the built-in Zachary karate club graph")
+    ap.add_argument("--lfr", metavar="NV", type=int, default=0,
+                    help="generate an LFR-style benchmark with NV vertices "
+                    "and planted ground-truth communities")
+    ap.add_argument("--mu", type=float, default=0.4,
+                    help="LFR mixing parameter")
     ap.add_argument("--device", choices=["auto", "cpu", "cuda"],
                     default="auto")
     ap.add_argument("--backend", choices=["auto", "hip", "torch"],
@@ -102,10 +108,11 @@ def validate(args, world: int):
     if args.early_term in (2, 4) and not 0.0 <= args.et_alpha <= 1.0:
         sys.exit("-a must be in [0,1]")
     n_sources = sum(bool(x) for x in
-                    (args.input, args.gen_nv, args.rmat, args.karate))
+                    (args.input, args.gen_nv, args.rmat, args.karate,
+                     args.lfr))
     if n_sources != 1:
         sys.exit("Specify exactly one graph source: -f FILE, -n NV, "
-                 "--rmat SCALE, or --karate")
+                 "--rmat SCALE, --lfr NV, or --karate")
     if args.random_edge_percent and not args.gen_nv:
         sys.exit("-e needs -n (generated graph)")
     if args.gen_nv and args.gen_nv % world != 0:
@@ -138,6 +145,11 @@ def _ingest(args, comm: Comm) -> DistGraph:
         if args.gen_out:
             write_dist_graph_collect(args.gen_out, dg, comm)
         return dg.to(dev)
+    if args.lfr:
+        dg, truth = lfr_dist_graph(args.lfr, comm.rank, comm.world,
+                                   mu=args.mu, seed=args.seed, device=dev)
+        args._lfr_truth = truth
+        return dg
     # --rmat
     return rmat_dist_graph(args.rmat, args.edgefactor, args.seed, comm, dev,
                            weight_dtype=wdtype)
@@ -218,7 +230,8 @@ def main(argv=None) -> int:
         print(f"Rebuild time: {res.times.get('rebuild', 0.0):.3f}s")
         print(f"Total time: {t_total:.3f}s  TEPS: {teps:.4g}")
 
-    if args.output or args.ground_truth:
+    builtin_truth = getattr(args, "_lfr_truth", None)
+    if args.output or args.ground_truth or builtin_truth is not None:
         # gather final communities to root in vertex order
         allc = comm.gather_cat(res.communities.to(comm.device), root=0)
         if comm.rank == 0:
@@ -228,9 +241,10 @@ def main(argv=None) -> int:
                     ".communities"
                 write_communities(out_path, allc)
                 print(f"Wrote communities to {out_path}")
-            if args.ground_truth:
-                truth = load_ground_truth(args.ground_truth,
-                                          zero_based=not args.one_based)
+            if args.ground_truth or builtin_truth is not None:
+                truth = builtin_truth if builtin_truth is not None else \
+                    load_ground_truth(args.ground_truth,
+                                      zero_based=not args.one_based)
                 m = compare_communities(truth, allc)
                 print(f"Ground truth: precision={m['precision']:.4f} "
                       f"recall={m['recall']:.4f} f-score={m['f_score']:.4f} "

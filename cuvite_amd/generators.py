@@ -18,6 +18,7 @@ MI355X-first):
 from __future__ import annotations
 
 import math
+from typing import Optional
 
 import numpy as np
 import torch
@@ -238,6 +239,133 @@ def rmat_graph(scale: int, edgefactor: int = 16, seed: int = 1,
     dst = torch.from_numpy(np.concatenate([v, u]))
     ww = torch.from_numpy(np.concatenate([w, w])).to(weight_dtype)
     return Graph.from_edge_tuples(nv, src, dst, ww)
+
+
+# ----------------------------------------------------------------- LFR -----
+
+def lfr_graph(nv: int, mu: float = 0.4, avg_deg: int = 20,
+              max_deg: Optional[int] = None, tau1: float = 2.5,
+              tau2: float = 1.5, min_comm: Optional[int] = None,
+              max_comm: Optional[int] = None, seed: int = 1,
+              weight_dtype: torch.dtype = torch.float64):
+    """LFR-style benchmark graph with planted communities.
+
+    The reference consumes LFR graphs as external files with a ground-truth
+    community file (-g, README:105-117; loadGroundTruthFile louvain.cpp:
+    3272-3304); this generator synthesizes an equivalent benchmark in-process
+    (no network access for the canonical LFR binaries): power-law degrees
+    (exponent tau1), power-law community sizes (exponent tau2), and mixing
+    parameter mu = fraction of each vertex's edges that leave its community.
+
+    Returns (Graph symmetrized, ground_truth int64 [nv]).
+    """
+    rng = np.random.Generator(np.random.Philox(key=[(seed << 32) | 0x1F12,
+                                                    nv & 0x7FFFFFFF]))
+    if max_deg is None:
+        max_deg = max(avg_deg * 5, int(math.sqrt(nv)))
+    min_deg = max(2, int(avg_deg * (tau1 - 2) / (tau1 - 1)))  # mean target
+    if min_comm is None:
+        min_comm = max(10, max_deg // 2)
+    if max_comm is None:
+        max_comm = max(min_comm + 1, nv // 10)
+
+    def powlaw(n, lo, hi, tau):
+        """n samples from P(x) ~ x^-tau on [lo, hi] (inverse transform)."""
+        u = rng.random(n)
+        a = 1.0 - tau
+        lo_a, hi_a = lo ** a, hi ** a
+        return np.floor((lo_a + u * (hi_a - lo_a)) ** (1.0 / a)).astype(
+            np.int64)
+
+    # community sizes until they cover nv
+    sizes = []
+    tot = 0
+    while tot < nv:
+        s = int(powlaw(1, min_comm, max_comm, tau2)[0])
+        sizes.append(s)
+        tot += s
+    sizes[-1] -= tot - nv  # trim the last one
+    if sizes[-1] < 2 and len(sizes) > 1:
+        sizes[-2] += sizes[-1]
+        sizes.pop()
+    sizes = np.array(sizes, dtype=np.int64)
+    ncomm = len(sizes)
+    truth = np.repeat(np.arange(ncomm, dtype=np.int64), sizes)
+    perm = rng.permutation(nv)
+    truth = truth[np.argsort(perm)]  # random vertex->community assignment
+
+    deg = powlaw(nv, min_deg, max_deg, tau1)
+    # cap intra-degree by community size - 1
+    comm_size_of = sizes[truth]
+    k_in = np.minimum(np.rint(deg * (1.0 - mu)).astype(np.int64),
+                      comm_size_of - 1)
+    k_out = deg - k_in
+
+    srcs, dsts = [], []
+
+    # intra-community configuration model per community
+    order = np.argsort(truth, kind="stable")
+    bounds = np.searchsorted(truth[order], np.arange(ncomm + 1))
+    for c in range(ncomm):
+        members = order[bounds[c]:bounds[c + 1]]
+        stubs = np.repeat(members, k_in[members])
+        if stubs.size < 2:
+            continue
+        rng.shuffle(stubs)
+        if stubs.size % 2:
+            stubs = stubs[:-1]
+        half = stubs.size // 2
+        a, b = stubs[:half], stubs[half:]
+        m = a != b
+        srcs.append(a[m])
+        dsts.append(b[m])
+
+    # inter-community stub pairing (reject pairs landing intra)
+    stubs = np.repeat(np.arange(nv, dtype=np.int64), k_out)
+    rng.shuffle(stubs)
+    if stubs.size % 2:
+        stubs = stubs[:-1]
+    half = stubs.size // 2
+    a, b = stubs[:half], stubs[half:]
+    m = (a != b) & (truth[a] != truth[b])
+    srcs.append(a[m])
+    dsts.append(b[m])
+
+    u = np.concatenate(srcs)
+    v = np.concatenate(dsts)
+    # dedup undirected pairs
+    lo_ = np.minimum(u, v)
+    hi_ = np.maximum(u, v)
+    key = lo_ * nv + hi_
+    key = np.unique(key)
+    u = key // nv
+    v = key % nv
+    w = rng.uniform(0.01, 1.0, u.size)
+
+    src = torch.from_numpy(np.concatenate([u, v]))
+    dst = torch.from_numpy(np.concatenate([v, u]))
+    ww = torch.from_numpy(np.concatenate([w, w])).to(weight_dtype)
+    g = Graph.from_edge_tuples(nv, src, dst, ww)
+    return g, torch.from_numpy(truth)
+
+
+def lfr_dist_graph(nv: int, rank: int, nranks: int, mu: float = 0.4,
+                   seed: int = 1, device=None, **kw):
+    """This rank's shard of an LFR benchmark (generated identically on every
+    rank from the seed, then sliced by the contiguous partition). LFR
+    acceptance configs are ~1M vertices — whole-graph generation per rank is
+    cheap next to 288 GB HBM. Returns (DistGraph, ground_truth_global)."""
+    from .graph import DistGraph, Graph as _G, Partition
+    g, truth = lfr_graph(nv, mu=mu, seed=seed, **kw)
+    part = Partition.contiguous(nv, nranks)
+    base, bound = part.base(rank), part.bound(rank)
+    e0, e1 = int(g.rowptr[base]), int(g.rowptr[bound])
+    lg = _G(g.rowptr[base:bound + 1] - g.rowptr[base],
+            g.tails[e0:e1], g.weights[e0:e1])
+    dg = DistGraph(lg, part, rank)
+    if device is not None:
+        dg = dg.to(device)
+    return dg, truth
 
 
 # ----------------------------------------------------------------- RGG -----

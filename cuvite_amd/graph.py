@@ -181,10 +181,21 @@ class DistGraph:
         return DistGraph(self.g.to(device), self.partition, self.rank)
 
     def ghost_vertices(self) -> torch.Tensor:
-        """Sorted unique global ids of remote tails (the rank's ghosts)."""
+        """Sorted unique global ids of remote tails (the rank's ghosts).
+        Chunked: torch advanced indexing overflows internal int32 offsets
+        past ~2^31 elements (observed as an absurd-size alloc at R-MAT s26)."""
         t = self.g.tails
-        remote = t[(t < self.base) | (t >= self.bound)]
-        return torch.unique(remote)
+        if self.partition.nranks == 1:
+            return torch.empty(0, dtype=torch.int64, device=t.device)
+        base, bound = self.base, self.bound
+        CH = 1 << 28
+        parts = []
+        for c0 in range(0, t.numel(), CH):
+            tc = t[c0:c0 + CH]
+            parts.append(torch.unique(tc[(tc < base) | (tc >= bound)]))
+        if not parts:
+            return torch.empty(0, dtype=torch.int64, device=t.device)
+        return torch.unique(torch.cat(parts))
 
     def local_degree_sum(self) -> torch.Tensor:
         """Per-vertex weighted degree (vDegree; ref louvain.cpp:2126-2151)."""

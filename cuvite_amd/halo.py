@@ -59,12 +59,19 @@ def build_halo(dg: DistGraph, comm: Comm) -> HaloContext:
     ng = ghosts.numel()
 
     # dense tails: local -> t - base ; ghost -> nv + position in sorted ghosts
-    is_local = (tails >= base) & (tails < bound)
-    dense = torch.empty_like(tails)
-    dense[is_local] = tails[is_local] - base
-    if ng:
-        dense[~is_local] = dg.nv + torch.searchsorted(ghosts, tails[~is_local])
-    tails_dense = dense.to(torch.int32)
+    # (chunked: advanced indexing overflows int32 internals past ~2^31 elems)
+    tails_dense = torch.empty(tails.numel(), dtype=torch.int32, device=dev)
+    CH = 1 << 28
+    for c0 in range(0, tails.numel(), CH):
+        tc = tails[c0:c0 + CH]
+        if ng == 0:
+            tails_dense[c0:c0 + CH] = (tc - base).to(torch.int32)
+            continue
+        is_local = (tc >= base) & (tc < bound)
+        dense = torch.where(
+            is_local, tc - base,
+            dg.nv + torch.searchsorted(ghosts, tc))
+        tails_dense[c0:c0 + CH] = dense.to(torch.int32)
 
     # ghosts grouped by owner (ghosts sorted => owner-contiguous segments)
     parts = dg.partition.parts.to(dev)
