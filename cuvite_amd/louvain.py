@@ -119,6 +119,12 @@ class PhaseState:
         comm_degree, comm_gid) with remote info fetched from owners."""
         dg, dev = self.dg, self.dg.g.device
         base, bound, nv = dg.base, dg.bound, dg.nv
+        if self.comm.world == 1:
+            # single-rank fast path: every label is local
+            dense = (self.curr_comm - base).to(torch.int32)
+            empty = torch.empty(0, dtype=torch.int64, device=dev)
+            return (dense, empty, self.local_size, self.local_degree,
+                    torch.arange(base, bound, device=dev))
         all_labels = torch.cat([self.curr_comm, ghost_comm])
         is_local = (all_labels >= base) & (all_labels < bound)
         remote_gids = torch.unique(all_labels[~is_local])
@@ -309,6 +315,10 @@ def _ordered_sweep(state: PhaseState, cfg: LouvainConfig, move_fn,
     dev = state.dg.g.device
     ghost_comm = exchange_ghost_labels(state.halo, state.curr_comm)
     dense, remote_gids, c_size, c_degree, c_gid = state.densify(ghost_comm)
+    # the per-class updates below mutate these; densify may return live
+    # references to state.local_size/local_degree (world-1 fast path)
+    c_size = c_size.clone()
+    c_degree = c_degree.clone()
     work_dense = dense.clone()
     cw_total = torch.zeros(nv, dtype=state.dg.g.weights.dtype, device=dev)
     for vidx in color_order:

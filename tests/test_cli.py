@@ -103,3 +103,33 @@ def test_ordering_matches_plain_quality():
         res = louvain(dg, Comm(torch.device("cpu")), cfg)
         qs[name] = res.modularity
     assert qs["order"] >= qs["plain"] - 0.05
+
+
+def test_lfr_generator_recovery():
+    """LFR planted communities are recovered at low mixing (acceptance-style
+    test, ref README:105-117 ground-truth workflow)."""
+    import torch
+    from cuvite_amd.generators import lfr_graph
+    from cuvite_amd.graph import single_partition
+    from cuvite_amd.louvain import louvain, LouvainConfig
+    from cuvite_amd.parallel import Comm
+    from cuvite_amd.compare import compare_communities
+
+    g, truth = lfr_graph(2000, mu=0.2, seed=5)
+    res = louvain(single_partition(g), Comm(torch.device("cpu")),
+                  LouvainConfig(backend="torch"))
+    m = compare_communities(truth, res.communities)
+    assert m["recall"] > 0.85
+    assert m["f_score"] > 0.8
+
+
+def test_lfr_dist_slices_cover_graph():
+    import torch
+    from cuvite_amd.generators import lfr_graph, lfr_dist_graph
+    g, truth = lfr_graph(1000, mu=0.4, seed=9)
+    ne = 0
+    for r in range(4):
+        dg, truth_r = lfr_dist_graph(1000, r, 4, mu=0.4, seed=9)
+        assert torch.equal(truth_r, truth)
+        ne += dg.ne
+    assert ne == g.ne
