@@ -437,13 +437,15 @@ __global__ __launch_bounds__(BLOCK) void row_sum_kernel(
     int64_t nv, W* __restrict__ out) {
   constexpr int WAVES = BLOCK / 64;
   const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  const int64_t v = (int64_t)blockIdx.x * WAVES + wave;
-  if (v >= nv) return;
-  const int64_t e0 = rowptr[v], e1 = rowptr[v + 1];
-  double acc = 0.0;
-  for (int64_t e = e0 + lane; e < e1; e += 64) acc += (double)weights[e];
-  acc = sum_reduce<64>(acc);
-  if (lane == 0) out[v] = (W)acc;
+  // wave-strided over vertices: grid stays under the 2^32-1 workitem cap
+  const int64_t vstride = (int64_t)gridDim.x * WAVES;
+  for (int64_t v = (int64_t)blockIdx.x * WAVES + wave; v < nv; v += vstride) {
+    const int64_t e0 = rowptr[v], e1 = rowptr[v + 1];
+    double acc = 0.0;
+    for (int64_t e = e0 + lane; e < e1; e += 64) acc += (double)weights[e];
+    acc = sum_reduce<64>(acc);
+    if (lane == 0) out[v] = (W)acc;
+  }
 }
 
 // ------------------------------- launchers ---------------------------------
@@ -593,7 +595,8 @@ void launch_row_sum(const int64_t* rowptr, const W* weights, int64_t nv,
   if (nv == 0) return;
   constexpr int BLOCK = 256;
   constexpr int WAVES = BLOCK / 64;
-  const int64_t grid = (nv + WAVES - 1) / WAVES;
+  int64_t grid = (nv + WAVES - 1) / WAVES;
+  if (grid > 1048576) grid = 1048576;  // stay under 2^32-1 workitems
   hipLaunchKernelGGL((row_sum_kernel<W, BLOCK>), dim3((uint32_t)grid),
                      dim3(BLOCK), 0, stream, rowptr, weights, nv, out);
 }
