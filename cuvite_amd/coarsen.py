@@ -99,12 +99,36 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor
     new_of_all = ref_new[torch.searchsorted(ref_uniq, all_cvect)]
 
     # --- 3. coarse edges: aggregate locally, route, merge, CSR ---------------
-    seg = torch.repeat_interleave(torch.arange(dg.nv, device=dev),
-                                  dg.g.degrees())
-    s_new = new_of_all[seg]
-    t_new = new_of_all[halo.tails_dense.to(torch.int64)]
-    w = dg.g.weights
-    s_agg, t_agg, w_agg = _aggregate(s_new, t_new, w, gnc)
+    # chunked over the edge list: torch sort/advanced-indexing are capped at
+    # INT_MAX elements and the per-chunk aggregates are far smaller than the
+    # fine edge list, so partial aggregation also bounds peak memory
+    ne = dg.g.ne
+    CH = 1 << 28
+    rowptr = dg.g.rowptr
+    parts_s, parts_t, parts_w = [], [], []
+    for c0 in range(0, max(ne, 1), CH):
+        c1 = min(c0 + CH, ne)
+        if c1 <= c0:
+            break
+        eidx = torch.arange(c0, c1, device=dev)
+        seg = torch.searchsorted(rowptr, eidx, right=True) - 1
+        del eidx
+        s_new = new_of_all[seg]
+        del seg
+        t_new = new_of_all[halo.tails_dense[c0:c1].to(torch.int64)]
+        cs, ct, cw = _aggregate(s_new, t_new, dg.g.weights[c0:c1], gnc)
+        parts_s.append(cs)
+        parts_t.append(ct)
+        parts_w.append(cw)
+    if parts_s:
+        s_agg, t_agg, w_agg = _aggregate(torch.cat(parts_s),
+                                         torch.cat(parts_t),
+                                         torch.cat(parts_w), gnc)
+    else:
+        s_agg = torch.zeros(0, dtype=torch.int64, device=dev)
+        t_agg = torch.zeros(0, dtype=torch.int64, device=dev)
+        w_agg = torch.zeros(0, dtype=dg.g.weights.dtype, device=dev)
+    del parts_s, parts_t, parts_w
 
     if world > 1:
         npdev = new_part.parts.to(dev)
