@@ -92,6 +92,9 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--threshold", type=float, default=1.0e-6)
     ap.add_argument("--max-phases", type=int, default=0,
                     help="cap the number of phases (0 = reference default)")
+    ap.add_argument("--diag-files", action="store_true",
+                    help="write per-rank diagnostics to dat.out.<rank> "
+                    "instead of stdout (ref main.cpp:101-110)")
     ap.add_argument("--stats", action="store_true",
                     help="print the graph distribution table "
                     "(ref printStats, distgraph.hpp:100-149)")
@@ -183,6 +186,9 @@ def main(argv=None) -> int:
     args = build_parser().parse_args(argv)
     comm = init_from_env(prefer=args.device)
     validate(args, comm.world)
+    if args.diag_files:
+        # reference default: per-rank dat.out.<rank> diagnostic files
+        sys.stdout = open(f"dat.out.{comm.rank}", "w")
 
     t0 = time.perf_counter()
     dg = _ingest(args, comm)
