@@ -71,6 +71,9 @@ def _buckets_for(rowptr: torch.Tensor):
     return vlists, offsets
 
 
+_pool_cache: dict = {}
+
+
 def local_move(inp):
     """HIP local-move iteration (see local_move.MoveInputs for semantics).
     Returns (target dense comm ids [nv], cluster_weight [nv])."""
@@ -78,8 +81,20 @@ def local_move(inp):
     vlists, offsets = _buckets_for(inp.rowptr)
     n_pool = int(offsets[-1])
     dev = inp.rowptr.device
-    pool_keys = torch.full((n_pool,), -1, dtype=torch.int32, device=dev)
-    pool_vals = torch.zeros(n_pool, dtype=inp.weights.dtype, device=dev)
+    # hub-table pool is phase-static: allocate once, reset per iteration
+    pk = (inp.rowptr.data_ptr(), n_pool, inp.weights.dtype)
+    hit = _pool_cache.get(pk)
+    if hit is None:
+        if len(_pool_cache) > 4:
+            _pool_cache.clear()
+        pool_keys = torch.empty(n_pool, dtype=torch.int32, device=dev)
+        pool_vals = torch.empty(n_pool, dtype=inp.weights.dtype, device=dev)
+        _pool_cache[pk] = (pool_keys, pool_vals)
+    else:
+        pool_keys, pool_vals = hit
+    if n_pool:
+        pool_keys.fill_(-1)
+        pool_vals.zero_()
     target, cw = ext.local_move_bucketed(
         inp.rowptr, inp.tails, inp.weights, inp.curr_comm, inp.v_degree,
         inp.comm_size, inp.comm_degree, inp.comm_gid, float(inp.constant),
