@@ -166,7 +166,8 @@ std::vector<at::Tensor> csr_from_edges(int64_t nv, int64_t base,
   cuvite::launch_degree_count(src.data_ptr<int64_t>(), ne, base,
                               cnt.data_ptr<int32_t>(), stream);
   auto rowptr = at::zeros({nv + 1}, src.options());
-  rowptr.narrow(0, 1, nv).copy_(at::cumsum(cnt, 0));
+  // int64 accumulation: an int32 cumsum overflows past 2^31 total edges
+  rowptr.narrow(0, 1, nv).copy_(at::cumsum(cnt, 0, at::kLong));
   auto tails = at::empty({ne}, src.options());
   auto weights = at::empty({ne}, w.options());
   cnt.zero_();  // reuse as the per-row placement cursor

@@ -138,3 +138,36 @@ def test_distributed_coloring_valid_and_matches_single():
     c2 = torch.cat([o[0] for o in outs])
     assert outs[0][1] == nc1
     assert torch.equal(c1, c2), "coloring must be P-independent"
+
+
+def _w_rmat_route(rank, world):
+    from cuvite_amd.generators import rmat_dist_graph
+    comm = Comm()
+    dg = rmat_dist_graph(9, 8, 4, comm, torch.device("cpu"))
+    return (dg.g.rowptr.cpu(), dg.g.tails.cpu(), dg.g.weights.cpu(), dg.base)
+
+
+def test_rmat_dist_routing_matches_single():
+    """Chunked owner-routing produces exactly the P=1 graph on any P."""
+    from cuvite_amd.generators import rmat_dist_graph
+
+    class _Solo:
+        world, rank, active = 1, 0, False
+        device = torch.device("cpu")
+        def allreduce_scalar(self, x, op="sum"):
+            return x
+        def all_to_all_v(self, send, recv_counts=None):
+            return [send[0]]
+    single = rmat_dist_graph(9, 8, 4, _Solo(), torch.device("cpu"))
+    outs = run_dist(2, _w_rmat_route)
+    rowptr = torch.cat([outs[0][0], outs[1][0][1:] + outs[0][0][-1]])
+    assert torch.equal(rowptr, single.g.rowptr)
+    # per-row multisets must match (row order within a rank may differ)
+    tails = torch.cat([outs[0][1], outs[1][1]])
+    weights = torch.cat([outs[0][2], outs[1][2]])
+    for v in range(0, 512, 23):
+        e0, e1 = int(single.g.rowptr[v]), int(single.g.rowptr[v + 1])
+        a = sorted(zip(tails[e0:e1].tolist(), weights[e0:e1].tolist()))
+        b = sorted(zip(single.g.tails[e0:e1].tolist(),
+                       single.g.weights[e0:e1].tolist()))
+        assert a == b
