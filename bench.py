@@ -22,6 +22,11 @@ import os
 import sys
 import time
 
+
+def _p(msg):
+    print(f"[bench {time.strftime('%H:%M:%S')}] {msg}", file=sys.stderr,
+          flush=True)
+
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -50,13 +55,20 @@ def main():
     wdtype = torch.float64 if args.dtype == "fp64" else torch.float32
 
     t_gen0 = time.perf_counter()
+    _p(f"generating rmat s{args.scale} on {device}")
     dg = rmat_dist_graph(args.scale, args.edgefactor, args.seed, comm,
                          device, weight_dtype=wdtype)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    _p(f"graph built: nv_local={dg.nv} ne_local={dg.ne}")
     ne_global = float(comm.allreduce_scalar(float(dg.ne)))
     t_gen = time.perf_counter() - t_gen0
 
     cfg = LouvainConfig(backend=args.backend)
     state = PhaseState(dg, comm)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    _p("phase state + halo ready")
     state.use_hip = device.type == "cuda" and args.backend in ("auto", "hip")
     move_fn = _pick_move_fn(cfg, device)
 
@@ -66,8 +78,11 @@ def main():
         state.past_comm, state.curr_comm = state.curr_comm, target
         return q
 
-    for _ in range(args.warmup):
+    for i in range(args.warmup):
         q = step()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        _p(f"warmup {i} done q={q:.6f}")
 
     comm.barrier()
     if device.type == "cuda":
