@@ -181,12 +181,23 @@ def rmat_dist_graph(scale: int, edgefactor: int, seed: int, comm,
         u, v = torch.from_numpy(uu), torch.from_numpy(vv)
         w = torch.from_numpy(ww).to(weight_dtype)
 
+    import os as _os
+    import sys as _sys
+    _dbg = _os.environ.get("CUVITE_PROGRESS")
+
+    def _p(m):
+        if _dbg:
+            print(f"[rmat] {m}", file=_sys.stderr, flush=True)
+
     if comm.world == 1:
+        _p(f"edges generated: {u.numel()}")
         src = torch.cat([u, v])
         dst = torch.cat([v, u])
         ww = torch.cat([w, w])
         del u, v, w
+        _p("symmetrized; building CSR")
         g = Graph.from_edge_tuples(nv, src, dst, ww)
+        _p("CSR done")
         return DistGraph(g, part, comm.rank)
 
     parts_dev = part.parts.to(device)
