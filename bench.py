@@ -72,9 +72,17 @@ def main():
     state.use_hip = device.type == "cuda" and args.backend in ("auto", "hip")
     move_fn = _pick_move_fn(cfg, device)
 
+    dbg = os.environ.get("CUVITE_PROGRESS")
+
     def step():
         target = _one_sweep(state, cfg, move_fn, None)
+        if dbg and device.type == "cuda":
+            torch.cuda.synchronize()
+            _p("sweep done")
         q = _modularity(state)
+        if dbg and device.type == "cuda":
+            torch.cuda.synchronize()
+            _p("modularity done")
         state.past_comm, state.curr_comm = state.curr_comm, target
         return q
 

@@ -95,6 +95,34 @@ def local_move(inp):
     if n_pool:
         pool_keys.fill_(-1)
         pool_vals.zero_()
+    if os.environ.get("CUVITE_PROGRESS"):
+        import sys
+        import time
+        torch.cuda.synchronize()
+        sizes = [int(v.numel()) for v in vlists]
+        print(f"[move] classes {sizes} pool={n_pool}", file=sys.stderr,
+              flush=True)
+        outs = []
+        for i in range(5):
+            one = [v if j == i else v[:0] for j, v in enumerate(vlists)]
+            t0 = time.perf_counter()
+            outs.append(ext.local_move_bucketed(
+                inp.rowptr, inp.tails, inp.weights, inp.curr_comm,
+                inp.v_degree, inp.comm_size, inp.comm_degree, inp.comm_gid,
+                float(inp.constant), one, offsets, pool_keys, pool_vals))
+            torch.cuda.synchronize()
+            print(f"[move] class {i} n={sizes[i]} "
+                  f"{time.perf_counter() - t0:.3f}s", file=sys.stderr,
+                  flush=True)
+        # merge: each class wrote its own vertices; take per-class targets
+        target = outs[0][0]
+        cw = outs[0][1]
+        for i in range(1, 5):
+            vl = vlists[i].to(torch.int64)
+            if vl.numel():
+                target[vl] = outs[i][0][vl]
+                cw[vl] = outs[i][1][vl]
+        return target, cw
     target, cw = ext.local_move_bucketed(
         inp.rowptr, inp.tails, inp.weights, inp.curr_comm, inp.v_degree,
         inp.comm_size, inp.comm_degree, inp.comm_gid, float(inp.constant),
