@@ -44,7 +44,7 @@ def _buckets_for(rowptr: torch.Tensor):
     key = (rowptr.data_ptr(), rowptr.numel())
     hit = _bucket_cache.get(key)
     if hit is not None and hit[0] is rowptr:  # identity check: ptr reuse safe
-        return hit[1], hit[2]
+        return hit[1], hit[2], hit[3], hit[4]
     deg = rowptr[1:] - rowptr[:-1]
     b0, b1, b2, b3 = _CLASS_BOUNDS
     vlists = [
@@ -63,12 +63,18 @@ def _buckets_for(rowptr: torch.Tensor):
         offsets = torch.zeros(hubs.numel() + 1, dtype=torch.int64,
                               device=rowptr.device)
         offsets[1:] = torch.cumsum(caps, dim=0)
+        eoffs = torch.zeros(hubs.numel() + 1, dtype=torch.int64,
+                            device=rowptr.device)
+        eoffs[1:] = torch.cumsum(hdeg, dim=0)
+        total_hub_edges = int(eoffs[-1])
     else:
         offsets = torch.zeros(1, dtype=torch.int64, device=rowptr.device)
+        eoffs = torch.zeros(1, dtype=torch.int64, device=rowptr.device)
+        total_hub_edges = 0
     if len(_bucket_cache) > 8:
         _bucket_cache.clear()
-    _bucket_cache[key] = (rowptr, vlists, offsets)
-    return vlists, offsets
+    _bucket_cache[key] = (rowptr, vlists, offsets, eoffs, total_hub_edges)
+    return vlists, offsets, eoffs, total_hub_edges
 
 
 _pool_cache: dict = {}
@@ -78,7 +84,7 @@ def local_move(inp):
     """HIP local-move iteration (see local_move.MoveInputs for semantics).
     Returns (target dense comm ids [nv], cluster_weight [nv])."""
     ext = _require()
-    vlists, offsets = _buckets_for(inp.rowptr)
+    vlists, offsets, eoffs, n_hub_edges = _buckets_for(inp.rowptr)
     n_pool = int(offsets[-1])
     dev = inp.rowptr.device
     # hub-table pool is phase-static: allocate once, reset per iteration
@@ -109,7 +115,8 @@ def local_move(inp):
             outs.append(ext.local_move_bucketed(
                 inp.rowptr, inp.tails, inp.weights, inp.curr_comm,
                 inp.v_degree, inp.comm_size, inp.comm_degree, inp.comm_gid,
-                float(inp.constant), one, offsets, pool_keys, pool_vals))
+                float(inp.constant), one, offsets, eoffs,
+                n_hub_edges if i == 4 else 0, pool_keys, pool_vals))
             torch.cuda.synchronize()
             print(f"[move] class {i} n={sizes[i]} "
                   f"{time.perf_counter() - t0:.3f}s", file=sys.stderr,
@@ -126,7 +133,7 @@ def local_move(inp):
     target, cw = ext.local_move_bucketed(
         inp.rowptr, inp.tails, inp.weights, inp.curr_comm, inp.v_degree,
         inp.comm_size, inp.comm_degree, inp.comm_gid, float(inp.constant),
-        vlists, offsets, pool_keys, pool_vals)
+        vlists, offsets, eoffs, n_hub_edges, pool_keys, pool_vals)
     return target, cw
 
 

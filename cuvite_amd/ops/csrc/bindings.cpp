@@ -29,8 +29,10 @@ void launch_sub(const int32_t*, int, const MoveArgs<W>&, hipStream_t);
 template <typename W>
 void launch_block(const int32_t*, int, const MoveArgs<W>&, hipStream_t);
 template <typename W>
-void launch_global(const int32_t*, int, const int64_t*, int32_t*, W*,
-                   const MoveArgs<W>&, hipStream_t);
+void launch_hub(const int32_t*, int, const int64_t*, int64_t, const int64_t*,
+                int32_t*, W*, double*, int32_t*, double*, int64_t*, int32_t*,
+                const MoveArgs<W>&, hipStream_t);
+constexpr int HUB_SPLITS_HOST = 16;  // keep in sync with HUB_SPLITS
 template <typename W>
 void launch_modularity(const W*, const W*, int64_t, double*, hipStream_t);
 template <typename W>
@@ -77,7 +79,8 @@ std::vector<at::Tensor> local_move(
     at::Tensor curr_comm, at::Tensor v_degree, at::Tensor comm_size,
     at::Tensor comm_degree, at::Tensor comm_gid, double constant,
     std::vector<at::Tensor> vlists, at::Tensor global_offsets,
-    at::Tensor pool_keys, at::Tensor pool_vals) {
+    at::Tensor hub_eoffs, int64_t total_hub_edges, at::Tensor pool_keys,
+    at::Tensor pool_vals) {
   CHECK_DEV(rowptr); CHECK_CONT(rowptr);
   CHECK_DEV(tails); CHECK_CONT(tails);
   CHECK_DEV(weights); CHECK_CONT(weights);
@@ -108,12 +111,27 @@ std::vector<at::Tensor> local_move(
     if (vlists[3].numel())
       cuvite::launch_block<W>(vlists[3].data_ptr<int32_t>(),
                               (int)vlists[3].numel(), args, stream);
-    if (vlists[4].numel())
-      cuvite::launch_global<W>(vlists[4].data_ptr<int32_t>(),
-                               (int)vlists[4].numel(),
-                               global_offsets.data_ptr<int64_t>(),
-                               pool_keys.data_ptr<int32_t>(),
-                               pool_vals.data_ptr<W>(), args, stream);
+    if (vlists[4].numel()) {
+      const int nhub = (int)vlists[4].numel();
+      auto hub_self = at::zeros({nhub}, rowptr.options().dtype(at::kDouble));
+      auto overflow = at::zeros({1}, rowptr.options().dtype(at::kInt));
+      auto p_gain = at::empty({nhub * cuvite::HUB_SPLITS_HOST},
+                              rowptr.options().dtype(at::kDouble));
+      auto p_gid = at::empty({nhub * cuvite::HUB_SPLITS_HOST},
+                             rowptr.options().dtype(at::kLong));
+      auto p_dense = at::empty({nhub * cuvite::HUB_SPLITS_HOST},
+                               rowptr.options().dtype(at::kInt));
+      cuvite::launch_hub<W>(vlists[4].data_ptr<int32_t>(), nhub,
+                            hub_eoffs.data_ptr<int64_t>(), total_hub_edges,
+                            global_offsets.data_ptr<int64_t>(),
+                            pool_keys.data_ptr<int32_t>(),
+                            pool_vals.data_ptr<W>(),
+                            hub_self.data_ptr<double>(),
+                            overflow.data_ptr<int32_t>(),
+                            p_gain.data_ptr<double>(),
+                            p_gid.data_ptr<int64_t>(),
+                            p_dense.data_ptr<int32_t>(), args, stream);
+    }
   });
   C10_HIP_CHECK(hipGetLastError());
   return {target, cw};

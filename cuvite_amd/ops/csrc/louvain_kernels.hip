@@ -14,7 +14,8 @@
 //   class 1: deg in (16, 64]     one wave/vertex, 128-slot LDS table
 //   class 2: deg in (64, 512]    one wave/vertex, 1024-slot LDS table
 //   block  : deg in (512, 4096]  one 256-thread block/vertex, 8192-slot LDS
-//   global : deg > 4096          one block/vertex, global-memory table
+//   hub    : deg > 4096          edge-parallel 3-kernel pipeline over a
+//                                global-memory table pool (see below)
 //
 // All gain arithmetic is fp64 regardless of the weight dtype so trajectories
 // match the fp64 CPU oracle (tie-break on equal gains -> smaller GLOBAL id).
@@ -266,87 +267,162 @@ __global__ __launch_bounds__(BLOCK) void lv_move_block(
 }
 
 // ---------------------------------------------------------------------------
-// Global-table kernel: one block per vertex, hub vertices (deg > 4096);
-// open-addressing table in a global scratch pool sized 2*deg rounded to a
-// power of two per vertex (offsets/caps precomputed on the host side).
-// The 288 GB HBM makes this pool a per-call tensor, not a hard-coded array
-// (the reference hard-codes 2x2.7 GB scatter buffers, GpuGraph.cu:61-64).
+// Hub pipeline (deg > 4096): three edge/slot-parallel kernels instead of the
+// reference's one-block-per-vertex histogram (distGetMaxIndex_large_new,
+// louvain_cuda.cu:878-1022). A single R-MAT mega-hub (degree in the
+// millions) must not pin one CU: inserts are grid-strided over ALL hub
+// edges (binary search locates the owning hub), the argmax scan is split
+// into HUB_SPLITS blocks per hub with a small partial-reduction array, and
+// a finalize wave combines partials. Tables live in a global pool sized
+// 2*deg rounded to a power of two per hub (288 GB HBM: sized from the
+// graph, unlike the reference's hard-coded 2x2.7 GB buffers,
+// GpuGraph.cu:61-64). Every probe loop is bounded by cap as a hang guard
+// (sets `overflow` instead of spinning).
 // ---------------------------------------------------------------------------
 
-template <typename W, int BLOCK>
-__global__ __launch_bounds__(BLOCK) void lv_move_global(
+constexpr int HUB_SPLITS = 16;
+
+template <typename W>
+__global__ void hub_insert_kernel(
     const int32_t* __restrict__ vlist, int nlist,
-    const int64_t* __restrict__ offsets,  // [nlist+1] table offsets (slots)
+    const int64_t* __restrict__ eoffs,  // [nlist+1] hub-degree prefix
+    const int64_t* __restrict__ toffs,  // [nlist+1] table offsets
     int32_t* __restrict__ pool_keys, W* __restrict__ pool_vals,
     const int64_t* __restrict__ rowptr, const int32_t* __restrict__ tails,
     const W* __restrict__ weights, const int32_t* __restrict__ curr_comm,
-    const W* __restrict__ v_degree, const int64_t* __restrict__ comm_size,
-    const W* __restrict__ comm_degree, const int64_t* __restrict__ comm_gid,
-    double constant, int32_t* __restrict__ target,
-    W* __restrict__ cluster_weight) {
-  constexpr int WAVES = BLOCK / 64;
-  __shared__ double red_gain[WAVES];
-  __shared__ int64_t red_gid[WAVES];
-  __shared__ int32_t red_dense[WAVES];
-  __shared__ double red_self[WAVES];
-
-  const int32_t v = vlist[blockIdx.x];
-  const int tid = threadIdx.x;
-  const int wave = tid / 64, lane = tid % 64;
-  const int64_t toff = offsets[blockIdx.x];
-  const int cap = (int)(offsets[blockIdx.x + 1] - toff);
-  int32_t* keys = pool_keys + toff;
-  W* vals = pool_vals + toff;
-
-  const int64_t e0 = rowptr[v], e1 = rowptr[v + 1];
-  const int32_t cc = curr_comm[v];
-  double selfloop = 0.0;
-  for (int64_t e = e0 + tid; e < e1; e += BLOCK) {
-    int32_t t = tails[e];
-    W w = weights[e];
-    if (t == v) selfloop += (double)w;
-    // global-memory insert (L2 atomics)
-    int32_t k = curr_comm[t];
+    double* __restrict__ hub_self, int32_t* __restrict__ overflow) {
+  const int64_t total = eoffs[nlist];
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; e < total;
+       e += stride) {
+    int lo = 0, hi = nlist;  // invariant: eoffs[lo] <= e < eoffs[hi]
+    while (hi - lo > 1) {
+      const int mid = (lo + hi) >> 1;
+      if (eoffs[mid] <= e) lo = mid; else hi = mid;
+    }
+    const int32_t v = vlist[lo];
+    const int64_t ee = rowptr[v] + (e - eoffs[lo]);
+    const int32_t t = tails[ee];
+    const W w = weights[ee];
+    if (t == v) unsafeAtomicAdd(&hub_self[lo], (double)w);
+    const int64_t toff = toffs[lo];
+    const int cap = (int)(toffs[lo + 1] - toff);
+    int32_t* keys = pool_keys + toff;
+    W* vals = pool_vals + toff;
+    const int32_t k = curr_comm[t];
     uint32_t h = hash_u32((uint32_t)k) & (cap - 1);
-    while (true) {
-      int32_t old = atomicCAS((int*)&keys[h], EMPTY_KEY, k);
+    for (int probes = 0; ; ++probes) {
+      const int32_t old = atomicCAS((int*)&keys[h], EMPTY_KEY, k);
       if (old == EMPTY_KEY || old == k) {
         unsafeAtomicAdd(&vals[h], w);
         break;
       }
       h = (h + 1) & (cap - 1);
+      if (probes > cap) { atomicExch(overflow, 1); break; }
     }
   }
-  selfloop = sum_reduce<64>(selfloop);
-  if (lane == 0) red_self[wave] = selfloop;
-  __syncthreads();
-  double self_total = 0.0;
-#pragma unroll
-  for (int i = 0; i < WAVES; i++) self_total += red_self[i];
+}
 
-  W wcc = table_probe(keys, vals, cap, cc);
-  double eix = (double)wcc - self_total;
-  double vdeg = (double)v_degree[v];
-  double ax = (double)comm_degree[cc] - vdeg;
+// Guarded probe for the cluster-weight lookup (cc may be absent).
+template <typename W>
+DEV_INLINE W pool_probe(const int32_t* keys, const W* vals, int cap,
+                        int32_t k) {
+  uint32_t h = hash_u32((uint32_t)k) & (cap - 1);
+  for (int probes = 0; probes <= cap; ++probes) {
+    const int32_t kk = keys[h];
+    if (kk == k) return vals[h];
+    if (kk == EMPTY_KEY) return (W)0;
+    h = (h + 1) & (cap - 1);
+  }
+  return (W)0;
+}
+
+template <typename W, int BLOCK>
+__global__ __launch_bounds__(BLOCK) void hub_scan_kernel(
+    const int32_t* __restrict__ vlist, int nlist,
+    const int64_t* __restrict__ toffs, const int32_t* __restrict__ pool_keys,
+    const W* __restrict__ pool_vals, const int32_t* __restrict__ curr_comm,
+    const W* __restrict__ v_degree, const W* __restrict__ comm_degree,
+    const int64_t* __restrict__ comm_gid, const double* __restrict__ hub_self,
+    double constant, double* __restrict__ p_gain,
+    int64_t* __restrict__ p_gid, int32_t* __restrict__ p_dense) {
+  constexpr int WAVES = BLOCK / 64;
+  __shared__ double red_gain[WAVES];
+  __shared__ int64_t red_gid[WAVES];
+  __shared__ int32_t red_dense[WAVES];
+
+  const int hidx = blockIdx.x / HUB_SPLITS;
+  const int part = blockIdx.x % HUB_SPLITS;
+  const int32_t v = vlist[hidx];
+  const int32_t cc = curr_comm[v];
+  const int64_t toff = toffs[hidx];
+  const int64_t cap = toffs[hidx + 1] - toff;
+  const int32_t* keys = pool_keys + toff;
+  const W* vals = pool_vals + toff;
+
+  const W wcc = pool_probe(keys, vals, (int)cap, cc);
+  const double eix = (double)wcc - hub_self[hidx];
+  const double vdeg = (double)v_degree[v];
+  const double ax = (double)comm_degree[cc] - vdeg;
+
+  const int64_t s0 = part * cap / HUB_SPLITS;
+  const int64_t s1 = (part + 1) * cap / HUB_SPLITS;
   Best best{0.0, comm_gid[cc], cc};
-  scan_slots(keys, vals, cap, tid, BLOCK, cc, eix, ax, vdeg, constant,
-             comm_degree, comm_gid, best);
+  for (int64_t s = s0 + threadIdx.x; s < s1; s += BLOCK) {
+    const int32_t y = keys[s];
+    if (y == EMPTY_KEY || y == cc) continue;
+    const double eiy = (double)vals[s];
+    const double ay = (double)comm_degree[y];
+    const double g = 2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant;
+    best_combine(best, g, comm_gid[y], y);
+  }
   best_reduce<64>(best);
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
   if (lane == 0) {
     red_gain[wave] = best.gain;
     red_gid[wave] = best.gid;
     red_dense[wave] = best.dense;
   }
   __syncthreads();
-  if (tid == 0) {
+  if (threadIdx.x == 0) {
 #pragma unroll
     for (int i = 1; i < WAVES; i++)
       best_combine(best, red_gain[i], red_gid[i], red_dense[i]);
+    p_gain[blockIdx.x] = best.gain;
+    p_gid[blockIdx.x] = best.gid;
+    p_dense[blockIdx.x] = best.dense;
+  }
+}
+
+template <typename W, int BLOCK>
+__global__ __launch_bounds__(BLOCK) void hub_finalize_kernel(
+    const int32_t* __restrict__ vlist, int nlist,
+    const int64_t* __restrict__ toffs, const int32_t* __restrict__ pool_keys,
+    const W* __restrict__ pool_vals, const int32_t* __restrict__ curr_comm,
+    const int64_t* __restrict__ comm_size, const int64_t* __restrict__ comm_gid,
+    const double* __restrict__ p_gain, const int64_t* __restrict__ p_gid,
+    const int32_t* __restrict__ p_dense, int32_t* __restrict__ target,
+    W* __restrict__ cluster_weight) {
+  constexpr int WAVES = BLOCK / 64;
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const int hidx = blockIdx.x * WAVES + wave;
+  if (hidx >= nlist) return;
+  const int32_t v = vlist[hidx];
+  const int32_t cc = curr_comm[v];
+  Best best{0.0, comm_gid[cc], cc};
+  if (lane < HUB_SPLITS) {
+    const int p = hidx * HUB_SPLITS + lane;
+    best = Best{p_gain[p], p_gid[p], p_dense[p]};
+  }
+  best_reduce<64>(best);
+  if (lane == 0) {
     int32_t tgt = best.dense;
     if (comm_size[tgt] == 1 && comm_size[cc] == 1 && best.gid > comm_gid[cc])
       tgt = cc;
     target[v] = tgt;
-    cluster_weight[v] = wcc;
+    cluster_weight[v] = pool_probe(pool_keys + toffs[hidx],
+                                   pool_vals + toffs[hidx],
+                                   (int)(toffs[hidx + 1] - toffs[hidx]), cc);
   }
 }
 
@@ -450,6 +526,13 @@ __global__ __launch_bounds__(BLOCK) void row_sum_kernel(
 
 // ------------------------------- launchers ---------------------------------
 
+static int grid_for(int64_t n, int block) {
+  int64_t g = (n + block - 1) / block;
+  // >> 2048 workgroups keeps all 8 XCDs fed; cap to bound tail latency
+  return (int)(g < 1 ? 1 : (g > 65535 ? 65535 : g));
+}
+
+
 template <typename W>
 struct MoveArgs {
   const int64_t* rowptr;
@@ -501,15 +584,28 @@ void launch_block(const int32_t* vlist, int nlist, const MoveArgs<W>& a,
 }
 
 template <typename W>
-void launch_global(const int32_t* vlist, int nlist, const int64_t* offsets,
-                   int32_t* pool_keys, W* pool_vals, const MoveArgs<W>& a,
-                   hipStream_t stream) {
-  constexpr int BLOCK = 512;
-  hipLaunchKernelGGL((lv_move_global<W, BLOCK>), dim3(nlist), dim3(BLOCK), 0,
-                     stream, vlist, nlist, offsets, pool_keys, pool_vals,
-                     a.rowptr, a.tails, a.weights, a.curr_comm, a.v_degree,
-                     a.comm_size, a.comm_degree, a.comm_gid, a.constant,
-                     a.target, a.cluster_weight);
+void launch_hub(const int32_t* vlist, int nlist, const int64_t* eoffs,
+                int64_t total_edges, const int64_t* toffs, int32_t* pool_keys,
+                W* pool_vals, double* hub_self, int32_t* overflow,
+                double* p_gain, int64_t* p_gid, int32_t* p_dense,
+                const MoveArgs<W>& a, hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  hipLaunchKernelGGL((hub_insert_kernel<W>),
+                     dim3(grid_for(total_edges, BLOCK)), dim3(BLOCK), 0,
+                     stream, vlist, nlist, eoffs, toffs, pool_keys, pool_vals,
+                     a.rowptr, a.tails, a.weights, a.curr_comm, hub_self,
+                     overflow);
+  hipLaunchKernelGGL((hub_scan_kernel<W, BLOCK>),
+                     dim3(nlist * HUB_SPLITS), dim3(BLOCK), 0, stream, vlist,
+                     nlist, toffs, pool_keys, pool_vals, a.curr_comm,
+                     a.v_degree, a.comm_degree, a.comm_gid, hub_self,
+                     a.constant, p_gain, p_gid, p_dense);
+  constexpr int WAVES = BLOCK / 64;
+  hipLaunchKernelGGL((hub_finalize_kernel<W, BLOCK>),
+                     dim3((nlist + WAVES - 1) / WAVES), dim3(BLOCK), 0,
+                     stream, vlist, nlist, toffs, pool_keys, pool_vals,
+                     a.curr_comm, a.comm_size, a.comm_gid, p_gain, p_gid,
+                     p_dense, a.target, a.cluster_weight);
 }
 
 // explicit instantiations used by bindings.cpp
@@ -522,9 +618,10 @@ void launch_global(const int32_t* vlist, int nlist, const int64_t* offsets,
                                         const MoveArgs<W>&, hipStream_t);    \
   template void launch_block<W>(const int32_t*, int, const MoveArgs<W>&,     \
                                 hipStream_t);                                 \
-  template void launch_global<W>(const int32_t*, int, const int64_t*,        \
-                                 int32_t*, W*, const MoveArgs<W>&,           \
-                                 hipStream_t);
+  template void launch_hub<W>(const int32_t*, int, const int64_t*, int64_t, \
+                              const int64_t*, int32_t*, W*, double*,         \
+                              int32_t*, double*, int64_t*, int32_t*,         \
+                              const MoveArgs<W>&, hipStream_t);
 
 INSTANTIATE(float)
 INSTANTIATE(double)
@@ -544,12 +641,6 @@ template void launch_modularity<float>(const float*, const float*, int64_t,
                                        double*, hipStream_t);
 template void launch_modularity<double>(const double*, const double*, int64_t,
                                         double*, hipStream_t);
-
-static int grid_for(int64_t n, int block) {
-  int64_t g = (n + block - 1) / block;
-  // >> 2048 workgroups keeps all 8 XCDs fed; cap to bound tail latency
-  return (int)(g < 1 ? 1 : (g > 65535 ? 65535 : g));
-}
 
 template <typename W>
 void launch_scatter_add(W* out, const int64_t* idx, const W* val, int64_t n,
