@@ -282,33 +282,36 @@ __global__ __launch_bounds__(BLOCK) void lv_move_block(
 
 constexpr int HUB_SPLITS = 16;
 
-template <typename W>
-__global__ void hub_insert_kernel(
+template <typename W, int BLOCK>
+__global__ __launch_bounds__(BLOCK) void hub_insert_kernel(
     const int32_t* __restrict__ vlist, int nlist,
-    const int64_t* __restrict__ eoffs,  // [nlist+1] hub-degree prefix
     const int64_t* __restrict__ toffs,  // [nlist+1] table offsets
     int32_t* __restrict__ pool_keys, W* __restrict__ pool_vals,
     const int64_t* __restrict__ rowptr, const int32_t* __restrict__ tails,
     const W* __restrict__ weights, const int32_t* __restrict__ curr_comm,
     double* __restrict__ hub_self, int32_t* __restrict__ overflow) {
-  const int64_t total = eoffs[nlist];
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; e < total;
-       e += stride) {
-    int lo = 0, hi = nlist;  // invariant: eoffs[lo] <= e < eoffs[hi]
-    while (hi - lo > 1) {
-      const int mid = (lo + hi) >> 1;
-      if (eoffs[mid] <= e) lo = mid; else hi = mid;
-    }
-    const int32_t v = vlist[lo];
-    const int64_t ee = rowptr[v] + (e - eoffs[lo]);
-    const int32_t t = tails[ee];
-    const W w = weights[ee];
-    if (t == v) unsafeAtomicAdd(&hub_self[lo], (double)w);
-    const int64_t toff = toffs[lo];
-    const int cap = (int)(toffs[lo + 1] - toff);
-    int32_t* keys = pool_keys + toff;
-    W* vals = pool_vals + toff;
+  // HUB_SPLITS blocks per hub, each inserting a contiguous slice of the
+  // hub's adjacency: bounds the number of blocks contending on one table
+  // (a full-grid stride sprays atomics from every XCD onto one table and
+  // measured ~40x slower on a degree-2^19 hub)
+  constexpr int WAVES = BLOCK / 64;
+  __shared__ double red_self[WAVES];
+  const int hidx = blockIdx.x / HUB_SPLITS;
+  const int part = blockIdx.x % HUB_SPLITS;
+  const int32_t v = vlist[hidx];
+  const int64_t e0 = rowptr[v], e1 = rowptr[v + 1];
+  const int64_t deg = e1 - e0;
+  const int64_t s0 = e0 + part * deg / HUB_SPLITS;
+  const int64_t s1 = e0 + (part + 1) * deg / HUB_SPLITS;
+  const int64_t toff = toffs[hidx];
+  const int cap = (int)(toffs[hidx + 1] - toff);
+  int32_t* keys = pool_keys + toff;
+  W* vals = pool_vals + toff;
+  double selfloop = 0.0;
+  for (int64_t e = s0 + threadIdx.x; e < s1; e += BLOCK) {
+    const int32_t t = tails[e];
+    const W w = weights[e];
+    if (t == v) selfloop += (double)w;
     const int32_t k = curr_comm[t];
     uint32_t h = hash_u32((uint32_t)k) & (cap - 1);
     for (int probes = 0; ; ++probes) {
@@ -320,6 +323,16 @@ __global__ void hub_insert_kernel(
       h = (h + 1) & (cap - 1);
       if (probes > cap) { atomicExch(overflow, 1); break; }
     }
+  }
+  selfloop = sum_reduce<64>(selfloop);
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  if (lane == 0) red_self[wave] = selfloop;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double tot = 0.0;
+#pragma unroll
+    for (int i = 0; i < WAVES; i++) tot += red_self[i];
+    if (tot != 0.0) unsafeAtomicAdd(&hub_self[hidx], tot);
   }
 }
 
@@ -590,9 +603,10 @@ void launch_hub(const int32_t* vlist, int nlist, const int64_t* eoffs,
                 double* p_gain, int64_t* p_gid, int32_t* p_dense,
                 const MoveArgs<W>& a, hipStream_t stream) {
   constexpr int BLOCK = 256;
-  hipLaunchKernelGGL((hub_insert_kernel<W>),
-                     dim3(grid_for(total_edges, BLOCK)), dim3(BLOCK), 0,
-                     stream, vlist, nlist, eoffs, toffs, pool_keys, pool_vals,
+  (void)eoffs; (void)total_edges;
+  hipLaunchKernelGGL((hub_insert_kernel<W, BLOCK>),
+                     dim3(nlist * HUB_SPLITS), dim3(BLOCK), 0,
+                     stream, vlist, nlist, toffs, pool_keys, pool_vals,
                      a.rowptr, a.tails, a.weights, a.curr_comm, hub_self,
                      overflow);
   hipLaunchKernelGGL((hub_scan_kernel<W, BLOCK>),
