@@ -80,27 +80,117 @@ def _buckets_for(rowptr: torch.Tensor):
 _pool_cache: dict = {}
 
 
+def _hub_moves_sorted(inp, hubs, hdeg, target, cw):
+    """Hub vertices (deg > 4096) via radix sort + segmented reduction instead
+    of the global hash-table pipeline: rocPRIM sort (torch.sort) of
+    (hub, community) keys, cumsum segment sums, and a vectorized exact-
+    tie-break argmax. Deterministic, atomic-free; the hash-table pipeline
+    (CUVITE_HUB_HIP=1) showed pathological slowdowns on tables past the 4 MB
+    XCD-L2 footprint (see profiles/ hang bisection)."""
+    dev = inp.rowptr.device
+    nhub = hubs.numel()
+    offs = torch.zeros(nhub + 1, dtype=torch.int64, device=dev)
+    offs[1:] = torch.cumsum(hdeg, dim=0)
+    tot = int(offs[-1])
+    seg = torch.repeat_interleave(torch.arange(nhub, device=dev), hdeg)
+    pos = torch.arange(tot, device=dev) - offs[seg]
+    eidx = inp.rowptr[hubs][seg] + pos
+    del pos
+    tails_h = inp.tails[eidx].to(torch.int64)
+    wts = inp.weights[eidx].to(torch.float64)
+    del eidx
+    comm = inp.curr_comm[tails_h].to(torch.int64)
+    v_of = hubs[seg]
+    selfmask = tails_h == v_of
+    del tails_h
+    # per-hub self-loop weight
+    selfloop = torch.zeros(nhub, dtype=torch.float64, device=dev)
+    if bool(selfmask.any()):
+        selfloop.index_add_(0, seg[selfmask], wts[selfmask])
+    C = inp.comm_degree.numel()
+    key = seg * C + comm
+    del comm
+    key_s, order = torch.sort(key)
+    del key
+    w_s = wts[order]
+    del wts, order
+    uniq, counts = torch.unique_consecutive(key_s, return_counts=True)
+    del key_s
+    ends = torch.cumsum(counts, dim=0) - 1
+    del counts
+    cs = torch.cumsum(w_s, dim=0)
+    del w_s
+    wsum = cs[ends].clone()
+    wsum[1:] -= cs[ends[:-1]]
+    del cs, ends
+    hub_of = uniq // C
+    y = uniq % C
+    del uniq
+
+    cc = inp.curr_comm[hubs].to(torch.int64)         # [nhub]
+    vdeg = inp.v_degree[hubs].to(torch.float64)
+    # weight to own community (includes self-loops)
+    is_cc = y == cc[hub_of]
+    wcc = torch.zeros(nhub, dtype=torch.float64, device=dev)
+    wcc[hub_of[is_cc]] = wsum[is_cc]
+    eix = wcc - selfloop
+    ax = inp.comm_degree[cc].to(torch.float64) - vdeg
+    ay = inp.comm_degree[y].to(torch.float64)
+    g = 2.0 * (wsum - eix[hub_of]) - 2.0 * vdeg[hub_of] * \
+        (ay - ax[hub_of]) * float(inp.constant)
+    g = torch.where(is_cc, torch.full_like(g, -torch.inf), g)
+    del ay, wsum
+
+    gmax = torch.full((nhub,), 0.0, dtype=torch.float64, device=dev)
+    gmax.scatter_reduce_(0, hub_of, g, reduce="amax")
+    move = gmax > 0.0
+    winner = (g == gmax[hub_of]) & move[hub_of] & ~is_cc
+    gid_y = inp.comm_gid[y]
+    gid_win = torch.full((nhub,), torch.iinfo(torch.int64).max,
+                         dtype=torch.int64, device=dev)
+    gid_win.scatter_reduce_(0, hub_of[winner], gid_y[winner], reduce="amin")
+    final = winner & (gid_y == gid_win[hub_of])
+    tgt_dense = cc.clone()
+    tgt_dense[hub_of[final]] = y[final]
+    # singleton-swap guard (louvain.cpp:2238-2239)
+    guard = (inp.comm_size[tgt_dense] == 1) & (inp.comm_size[cc] == 1) & \
+        (inp.comm_gid[tgt_dense] > inp.comm_gid[cc])
+    tgt_dense = torch.where(guard, cc, tgt_dense)
+    target[hubs] = tgt_dense.to(torch.int32)
+    cw[hubs] = wcc.to(cw.dtype)
+
+
 def local_move(inp):
     """HIP local-move iteration (see local_move.MoveInputs for semantics).
     Returns (target dense comm ids [nv], cluster_weight [nv])."""
     ext = _require()
     vlists, offsets, eoffs, n_hub_edges = _buckets_for(inp.rowptr)
-    n_pool = int(offsets[-1])
     dev = inp.rowptr.device
-    # hub-table pool is phase-static: allocate once, reset per iteration
-    pk = (inp.rowptr.data_ptr(), n_pool, inp.weights.dtype)
-    hit = _pool_cache.get(pk)
-    if hit is None:
-        if len(_pool_cache) > 4:
-            _pool_cache.clear()
-        pool_keys = torch.empty(n_pool, dtype=torch.int32, device=dev)
-        pool_vals = torch.empty(n_pool, dtype=inp.weights.dtype, device=dev)
-        _pool_cache[pk] = (pool_keys, pool_vals)
+    hub_hip = bool(os.environ.get("CUVITE_HUB_HIP"))
+    if not hub_hip:
+        # hub class handled by the sort-based path after the HIP classes
+        n_pool = 0
+        pool_keys = torch.empty(0, dtype=torch.int32, device=dev)
+        pool_vals = torch.empty(0, dtype=inp.weights.dtype, device=dev)
+        hubs64 = vlists[4].to(torch.int64)
+        vlists = vlists[:4] + [vlists[4][:0]]
     else:
-        pool_keys, pool_vals = hit
-    if n_pool:
-        pool_keys.fill_(-1)
-        pool_vals.zero_()
+        n_pool = int(offsets[-1])
+        # hub-table pool is phase-static: allocate once, reset per iteration
+        pk = (inp.rowptr.data_ptr(), n_pool, inp.weights.dtype)
+        hit = _pool_cache.get(pk)
+        if hit is None:
+            if len(_pool_cache) > 4:
+                _pool_cache.clear()
+            pool_keys = torch.empty(n_pool, dtype=torch.int32, device=dev)
+            pool_vals = torch.empty(n_pool, dtype=inp.weights.dtype,
+                                    device=dev)
+            _pool_cache[pk] = (pool_keys, pool_vals)
+        else:
+            pool_keys, pool_vals = hit
+        if n_pool:
+            pool_keys.fill_(-1)
+            pool_vals.zero_()
     if os.environ.get("CUVITE_PROGRESS"):
         import sys
         import time
@@ -129,11 +219,22 @@ def local_move(inp):
             if vl.numel():
                 target[vl] = outs[i][0][vl]
                 cw[vl] = outs[i][1][vl]
+        if not hub_hip and hubs64.numel():
+            t0 = time.perf_counter()
+            hdeg = eoffs[1:] - eoffs[:-1]
+            _hub_moves_sorted(inp, hubs64, hdeg, target, cw)
+            torch.cuda.synchronize()
+            print(f"[move] hub-sort n={hubs64.numel()} "
+                  f"{time.perf_counter() - t0:.3f}s", file=sys.stderr,
+                  flush=True)
         return target, cw
     target, cw = ext.local_move_bucketed(
         inp.rowptr, inp.tails, inp.weights, inp.curr_comm, inp.v_degree,
         inp.comm_size, inp.comm_degree, inp.comm_gid, float(inp.constant),
         vlists, offsets, eoffs, n_hub_edges, pool_keys, pool_vals)
+    if not hub_hip and hubs64.numel():
+        hdeg = eoffs[1:] - eoffs[:-1]
+        _hub_moves_sorted(inp, hubs64, hdeg, target, cw)
     return target, cw
 
 

@@ -112,3 +112,44 @@ def test_modularity_parts():
     cd = torch.tensor([3.0, 4.0], dtype=torch.float64)
     p = modularity_parts(cw, cd)
     assert p.tolist() == [3.0, 25.0]
+
+
+def test_hub_moves_sorted_matches_oracle():
+    """The sort-based hub path (ops._hub_moves_sorted) must reproduce the
+    torch oracle exactly on unit-weight graphs, including tie-breaks and the
+    singleton guard. Runs on CPU (pure torch ops)."""
+    from cuvite_amd.generators import rmat_graph
+    from cuvite_amd import ops
+
+    torch.manual_seed(3)
+    g = rmat_graph(9, 24, seed=7)
+    g.weights.fill_(1.0)
+    nv = g.nv
+    for state in ("singleton", "random"):
+        if state == "singleton":
+            curr = torch.arange(nv, dtype=torch.int64)
+        else:
+            curr = torch.randint(0, nv, (nv,),
+                                 generator=torch.Generator().manual_seed(4))
+        size = torch.zeros(nv, dtype=torch.int64)
+        size.index_add_(0, curr, torch.ones(nv, dtype=torch.int64))
+        deg = g.rowptr[1:] - g.rowptr[:-1]
+        vdeg = torch.zeros(nv, dtype=torch.float64)
+        seg = torch.repeat_interleave(torch.arange(nv), deg)
+        vdeg.index_add_(0, seg, g.weights)
+        cdeg = torch.zeros(nv, dtype=torch.float64)
+        cdeg.index_add_(0, curr, vdeg)
+        inp = MoveInputs(g.rowptr, g.tails.to(torch.int32), g.weights,
+                         curr.to(torch.int32), vdeg, size, cdeg,
+                         torch.arange(nv, dtype=torch.int64),
+                         1.0 / float(vdeg.sum()))
+        t_ref, cw_ref = local_move_torch(inp)
+        # treat every vertex with deg > 64 as a "hub" for this test
+        hubs = (deg > 64).nonzero(as_tuple=True)[0]
+        assert hubs.numel() > 3, "need hubs for the test to be meaningful"
+        target = inp.curr_comm.clone()
+        cw = torch.zeros(nv, dtype=torch.float64)
+        ops._hub_moves_sorted(inp, hubs, deg[hubs], target, cw)
+        assert torch.equal(target[hubs].to(torch.int64),
+                           t_ref[hubs].to(torch.int64))
+        assert torch.allclose(cw[hubs], cw_ref[hubs])
