@@ -80,7 +80,25 @@ def _buckets_for(rowptr: torch.Tensor):
 _pool_cache: dict = {}
 
 
+_HUB_SORT_CHUNK = 1 << 30  # max edges per sort batch (torch op INT_MAX cap)
+
+
 def _hub_moves_sorted(inp, hubs, hdeg, target, cw):
+    """Chunk wrapper: split the hub list into groups whose edge totals stay
+    under the torch sort cap; each hub's candidates are independent."""
+    total = int(hdeg.sum())
+    if total <= _HUB_SORT_CHUNK or hubs.numel() <= 1:
+        return _hub_moves_sorted_one(inp, hubs, hdeg, target, cw)
+    cum = torch.cumsum(hdeg, dim=0)
+    group = torch.div(cum - 1, _HUB_SORT_CHUNK, rounding_mode="floor")
+    for gidx in range(int(group[-1]) + 1):
+        m = group == gidx
+        if bool(m.any()):
+            _hub_moves_sorted_one(inp, hubs[m], hdeg[m], target, cw)
+    return None
+
+
+def _hub_moves_sorted_one(inp, hubs, hdeg, target, cw):
     """Hub vertices (deg > 4096) via radix sort + segmented reduction instead
     of the global hash-table pipeline: rocPRIM sort (torch.sort) of
     (hub, community) keys, cumsum segment sums, and a vectorized exact-

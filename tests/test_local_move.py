@@ -153,3 +153,35 @@ def test_hub_moves_sorted_matches_oracle():
         assert torch.equal(target[hubs].to(torch.int64),
                            t_ref[hubs].to(torch.int64))
         assert torch.allclose(cw[hubs], cw_ref[hubs])
+
+
+def test_hub_moves_sorted_chunked_matches_unchunked():
+    from cuvite_amd.generators import rmat_graph
+    from cuvite_amd import ops
+
+    g = rmat_graph(8, 24, seed=2)
+    nv = g.nv
+    deg = g.rowptr[1:] - g.rowptr[:-1]
+    curr = torch.arange(nv, dtype=torch.int64)
+    vdeg = torch.zeros(nv, dtype=torch.float64)
+    seg = torch.repeat_interleave(torch.arange(nv), deg)
+    vdeg.index_add_(0, seg, g.weights)
+    inp = MoveInputs(g.rowptr, g.tails.to(torch.int32), g.weights,
+                     curr.to(torch.int32), vdeg,
+                     torch.ones(nv, dtype=torch.int64), vdeg.clone(),
+                     torch.arange(nv, dtype=torch.int64),
+                     1.0 / float(vdeg.sum()))
+    hubs = (deg > 32).nonzero(as_tuple=True)[0]
+    t1 = inp.curr_comm.clone()
+    cw1 = torch.zeros(nv, dtype=torch.float64)
+    ops._hub_moves_sorted(inp, hubs, deg[hubs], t1, cw1)
+    old = ops._HUB_SORT_CHUNK
+    try:
+        ops._HUB_SORT_CHUNK = 200  # force many groups
+        t2 = inp.curr_comm.clone()
+        cw2 = torch.zeros(nv, dtype=torch.float64)
+        ops._hub_moves_sorted(inp, hubs, deg[hubs], t2, cw2)
+    finally:
+        ops._HUB_SORT_CHUNK = old
+    assert torch.equal(t1, t2)
+    assert torch.allclose(cw1, cw2)
