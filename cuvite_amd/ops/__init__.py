@@ -78,27 +78,42 @@ def _buckets_for(rowptr: torch.Tensor):
 
 
 _pool_cache: dict = {}
+_side_streams: dict = {}
+
+
+def _side_stream(dev):
+    st = _side_streams.get(dev)
+    if st is None:
+        st = torch.cuda.Stream(device=dev)
+        _side_streams[dev] = st
+    return st
 
 
 _HUB_SORT_CHUNK = 1 << 30  # max edges per sort batch (torch op INT_MAX cap)
 
 
-def _hub_moves_sorted(inp, hubs, hdeg, target, cw):
+def _hub_moves_sorted(inp, hubs, hdeg):
     """Chunk wrapper: split the hub list into groups whose edge totals stay
-    under the torch sort cap; each hub's candidates are independent."""
+    under the torch sort cap; each hub's candidates are independent.
+    Returns (target_dense int32 [nhub], wcc [nhub]) aligned with `hubs`."""
     total = int(hdeg.sum())
     if total <= _HUB_SORT_CHUNK or hubs.numel() <= 1:
-        return _hub_moves_sorted_one(inp, hubs, hdeg, target, cw)
+        return _hub_moves_sorted_one(inp, hubs, hdeg)
     cum = torch.cumsum(hdeg, dim=0)
     group = torch.div(cum - 1, _HUB_SORT_CHUNK, rounding_mode="floor")
+    tgt = torch.empty(hubs.numel(), dtype=torch.int32, device=hubs.device)
+    wcc = torch.empty(hubs.numel(), dtype=inp.weights.dtype,
+                      device=hubs.device)
     for gidx in range(int(group[-1]) + 1):
         m = group == gidx
         if bool(m.any()):
-            _hub_moves_sorted_one(inp, hubs[m], hdeg[m], target, cw)
-    return None
+            t, w = _hub_moves_sorted_one(inp, hubs[m], hdeg[m])
+            tgt[m] = t
+            wcc[m] = w
+    return tgt, wcc
 
 
-def _hub_moves_sorted_one(inp, hubs, hdeg, target, cw):
+def _hub_moves_sorted_one(inp, hubs, hdeg):
     """Hub vertices (deg > 4096) via radix sort + segmented reduction instead
     of the global hash-table pipeline: rocPRIM sort (torch.sort) of
     (hub, community) keys, cumsum segment sums, and a vectorized exact-
@@ -174,8 +189,7 @@ def _hub_moves_sorted_one(inp, hubs, hdeg, target, cw):
     guard = (inp.comm_size[tgt_dense] == 1) & (inp.comm_size[cc] == 1) & \
         (inp.comm_gid[tgt_dense] > inp.comm_gid[cc])
     tgt_dense = torch.where(guard, cc, tgt_dense)
-    target[hubs] = tgt_dense.to(torch.int32)
-    cw[hubs] = wcc.to(cw.dtype)
+    return tgt_dense.to(torch.int32), wcc.to(inp.weights.dtype)
 
 
 def local_move(inp):
@@ -240,19 +254,47 @@ def local_move(inp):
         if not hub_hip and hubs64.numel():
             t0 = time.perf_counter()
             hdeg = eoffs[1:] - eoffs[:-1]
-            _hub_moves_sorted(inp, hubs64, hdeg, target, cw)
+            hub_tgt, hub_cw = _hub_moves_sorted(inp, hubs64, hdeg)
+            target[hubs64] = hub_tgt
+            cw[hubs64] = hub_cw
             torch.cuda.synchronize()
             print(f"[move] hub-sort n={hubs64.numel()} "
                   f"{time.perf_counter() - t0:.3f}s", file=sys.stderr,
                   flush=True)
         return target, cw
+    run_hub_sort = not hub_hip and hubs64.numel() > 0
+    overlap = run_hub_sort and not os.environ.get("CUVITE_NO_OVERLAP")
+    if overlap:
+        # disjoint vertex sets: run the HIP class kernels on a side stream
+        # concurrently with the (longer) hub sort on the main stream; hub
+        # results come back as separate tensors and are scattered after the
+        # join (no cross-stream writes into the class kernels' outputs)
+        main = torch.cuda.current_stream(dev)
+        side = _side_stream(dev)
+        side.wait_stream(main)
+        with torch.cuda.stream(side):
+            target, cw = ext.local_move_bucketed(
+                inp.rowptr, inp.tails, inp.weights, inp.curr_comm,
+                inp.v_degree, inp.comm_size, inp.comm_degree, inp.comm_gid,
+                float(inp.constant), vlists, offsets, eoffs, n_hub_edges,
+                pool_keys, pool_vals)
+        hdeg = eoffs[1:] - eoffs[:-1]
+        hub_tgt, hub_cw = _hub_moves_sorted(inp, hubs64, hdeg)
+        main.wait_stream(side)
+        target.record_stream(main)
+        cw.record_stream(main)
+        target[hubs64] = hub_tgt
+        cw[hubs64] = hub_cw
+        return target, cw
     target, cw = ext.local_move_bucketed(
         inp.rowptr, inp.tails, inp.weights, inp.curr_comm, inp.v_degree,
         inp.comm_size, inp.comm_degree, inp.comm_gid, float(inp.constant),
         vlists, offsets, eoffs, n_hub_edges, pool_keys, pool_vals)
-    if not hub_hip and hubs64.numel():
+    if run_hub_sort:
         hdeg = eoffs[1:] - eoffs[:-1]
-        _hub_moves_sorted(inp, hubs64, hdeg, target, cw)
+        hub_tgt, hub_cw = _hub_moves_sorted(inp, hubs64, hdeg)
+        target[hubs64] = hub_tgt
+        cw[hubs64] = hub_cw
     return target, cw
 
 

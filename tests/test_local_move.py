@@ -147,12 +147,10 @@ def test_hub_moves_sorted_matches_oracle():
         # treat every vertex with deg > 64 as a "hub" for this test
         hubs = (deg > 64).nonzero(as_tuple=True)[0]
         assert hubs.numel() > 3, "need hubs for the test to be meaningful"
-        target = inp.curr_comm.clone()
-        cw = torch.zeros(nv, dtype=torch.float64)
-        ops._hub_moves_sorted(inp, hubs, deg[hubs], target, cw)
-        assert torch.equal(target[hubs].to(torch.int64),
+        hub_tgt, hub_cw = ops._hub_moves_sorted(inp, hubs, deg[hubs])
+        assert torch.equal(hub_tgt.to(torch.int64),
                            t_ref[hubs].to(torch.int64))
-        assert torch.allclose(cw[hubs], cw_ref[hubs])
+        assert torch.allclose(hub_cw, cw_ref[hubs])
 
 
 def test_hub_moves_sorted_chunked_matches_unchunked():
@@ -172,15 +170,11 @@ def test_hub_moves_sorted_chunked_matches_unchunked():
                      torch.arange(nv, dtype=torch.int64),
                      1.0 / float(vdeg.sum()))
     hubs = (deg > 32).nonzero(as_tuple=True)[0]
-    t1 = inp.curr_comm.clone()
-    cw1 = torch.zeros(nv, dtype=torch.float64)
-    ops._hub_moves_sorted(inp, hubs, deg[hubs], t1, cw1)
+    t1, cw1 = ops._hub_moves_sorted(inp, hubs, deg[hubs])
     old = ops._HUB_SORT_CHUNK
     try:
         ops._HUB_SORT_CHUNK = 200  # force many groups
-        t2 = inp.curr_comm.clone()
-        cw2 = torch.zeros(nv, dtype=torch.float64)
-        ops._hub_moves_sorted(inp, hubs, deg[hubs], t2, cw2)
+        t2, cw2 = ops._hub_moves_sorted(inp, hubs, deg[hubs])
     finally:
         ops._HUB_SORT_CHUNK = old
     assert torch.equal(t1, t2)
