@@ -171,3 +171,44 @@ def test_rmat_dist_routing_matches_single():
         b = sorted(zip(single.g.tails[e0:e1].tolist(),
                        single.g.weights[e0:e1].tolist()))
         assert a == b
+
+
+def _w_louvain_balanced(rank, world):
+    g = rmat_graph(8, 8, seed=3)
+    part = Partition.edge_balanced(g.rowptr, world)
+    b, e = part.base(rank), part.bound(rank)
+    rp = g.rowptr[b:e + 1] - g.rowptr[b]
+    e0, e1 = int(g.rowptr[b]), int(g.rowptr[e])
+    dg = DistGraph(Graph(rp.clone(), g.tails[e0:e1].clone(),
+                         g.weights[e0:e1].clone()), part, rank)
+    res = louvain(dg, Comm(), LouvainConfig(backend="torch"))
+    return res.modularity, res.communities.cpu(), dg.nv
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_distributed_louvain_edge_balanced(world):
+    """Edge-balanced partition (-b) gives the same result as contiguous."""
+    single = louvain(single_partition(rmat_graph(8, 8, seed=3)), Comm(),
+                     LouvainConfig(backend="torch"))
+    outs = run_dist(world, _w_louvain_balanced)
+    comms = torch.cat([o[1] for o in outs])
+    assert sum(o[2] for o in outs) == 256
+    assert abs(outs[0][0] - single.modularity) < 1e-9
+    assert torch.equal(comms, single.communities)
+
+
+def test_distributed_louvain_ordering():
+    """-d (color-ordered, no per-class sync) across 2 ranks reaches
+    comparable modularity to the single-rank plain run."""
+    single = louvain(single_partition(karate_graph()), Comm(),
+                     LouvainConfig(backend="torch"))
+    outs = run_dist(2, _w_louvain, "karate",
+                    {"ordering": True, "max_colors": 6})
+    assert outs[0][0] >= single.modularity - 0.06
+
+
+@pytest.mark.parametrize("et", [1, 2, 3, 4])
+def test_distributed_louvain_early_term(et):
+    outs = run_dist(2, _w_louvain, "karate",
+                    {"early_term": et, "et_delta": 0.5})
+    assert outs[0][0] > 0.15  # converges to a sane modularity (ET trades quality)
