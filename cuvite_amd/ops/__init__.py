@@ -44,7 +44,7 @@ def _buckets_for(rowptr: torch.Tensor):
     key = (rowptr.data_ptr(), rowptr.numel())
     hit = _bucket_cache.get(key)
     if hit is not None and hit[0] is rowptr:  # identity check: ptr reuse safe
-        return hit[1], hit[2], hit[3], hit[4]
+        return hit[1], hit[2], hit[3], hit[4], hit[5]
     deg = rowptr[1:] - rowptr[:-1]
     b0, b1, b2, b3 = _CLASS_BOUNDS
     vlists = [
@@ -71,10 +71,12 @@ def _buckets_for(rowptr: torch.Tensor):
         offsets = torch.zeros(1, dtype=torch.int64, device=rowptr.device)
         eoffs = torch.zeros(1, dtype=torch.int64, device=rowptr.device)
         total_hub_edges = 0
+    hubs64 = hubs.to(torch.int64)
     if len(_bucket_cache) > 8:
         _bucket_cache.clear()
-    _bucket_cache[key] = (rowptr, vlists, offsets, eoffs, total_hub_edges)
-    return vlists, offsets, eoffs, total_hub_edges
+    _bucket_cache[key] = (rowptr, vlists, offsets, eoffs, total_hub_edges,
+                          hubs64)
+    return vlists, offsets, eoffs, total_hub_edges, hubs64
 
 
 _pool_cache: dict = {}
@@ -113,14 +115,19 @@ def _hub_moves_sorted(inp, hubs, hdeg):
     return tgt, wcc
 
 
-def _hub_moves_sorted_one(inp, hubs, hdeg):
-    """Hub vertices (deg > 4096) via radix sort + segmented reduction instead
-    of the global hash-table pipeline: rocPRIM sort (torch.sort) of
-    (hub, community) keys, cumsum segment sums, and a vectorized exact-
-    tie-break argmax. Deterministic, atomic-free; the hash-table pipeline
-    (CUVITE_HUB_HIP=1) showed pathological slowdowns on tables past the 4 MB
-    XCD-L2 footprint (see profiles/ hang bisection)."""
+_hub_static_cache: dict = {}
+
+
+def _hub_static(inp, hubs, hdeg):
+    """Phase-static hub data: flattened hub adjacency (tails, weights),
+    per-hub segment ids and self-loop weights. None of it depends on the
+    community labels, so it is gathered once per phase and reused every
+    iteration (cache keyed like _bucket_cache)."""
     dev = inp.rowptr.device
+    key = (inp.rowptr.data_ptr(), hubs.data_ptr(), hubs.numel())
+    hit = _hub_static_cache.get(key)
+    if hit is not None and hit[0] is inp.rowptr and hit[1] is hubs:
+        return hit[2]
     nhub = hubs.numel()
     offs = torch.zeros(nhub + 1, dtype=torch.int64, device=dev)
     offs[1:] = torch.cumsum(hdeg, dim=0)
@@ -132,14 +139,29 @@ def _hub_moves_sorted_one(inp, hubs, hdeg):
     tails_h = inp.tails[eidx].to(torch.int64)
     wts = inp.weights[eidx].to(torch.float64)
     del eidx
-    comm = inp.curr_comm[tails_h].to(torch.int64)
-    v_of = hubs[seg]
-    selfmask = tails_h == v_of
-    del tails_h
-    # per-hub self-loop weight
+    selfmask = tails_h == hubs[seg]
     selfloop = torch.zeros(nhub, dtype=torch.float64, device=dev)
     if bool(selfmask.any()):
         selfloop.index_add_(0, seg[selfmask], wts[selfmask])
+    del selfmask
+    data = (seg, tails_h, wts, selfloop)
+    if len(_hub_static_cache) > 4:
+        _hub_static_cache.clear()
+    _hub_static_cache[key] = (inp.rowptr, hubs, data)
+    return data
+
+
+def _hub_moves_sorted_one(inp, hubs, hdeg):
+    """Hub vertices (deg > 4096) via radix sort + segmented reduction instead
+    of the global hash-table pipeline: rocPRIM sort (torch.sort) of
+    (hub, community) keys, cumsum segment sums, and a vectorized exact-
+    tie-break argmax. Deterministic, atomic-free; the hash-table pipeline
+    (CUVITE_HUB_HIP=1) showed pathological slowdowns on tables past the 4 MB
+    XCD-L2 footprint (see profiles/ hang bisection)."""
+    dev = inp.rowptr.device
+    nhub = hubs.numel()
+    seg, tails_h, wts, selfloop = _hub_static(inp, hubs, hdeg)
+    comm = inp.curr_comm[tails_h].to(torch.int64)
     C = inp.comm_degree.numel()
     key = seg * C + comm
     del comm
@@ -196,7 +218,7 @@ def local_move(inp):
     """HIP local-move iteration (see local_move.MoveInputs for semantics).
     Returns (target dense comm ids [nv], cluster_weight [nv])."""
     ext = _require()
-    vlists, offsets, eoffs, n_hub_edges = _buckets_for(inp.rowptr)
+    vlists, offsets, eoffs, n_hub_edges, hubs64 = _buckets_for(inp.rowptr)
     dev = inp.rowptr.device
     hub_hip = bool(os.environ.get("CUVITE_HUB_HIP"))
     if not hub_hip:
@@ -204,7 +226,6 @@ def local_move(inp):
         n_pool = 0
         pool_keys = torch.empty(0, dtype=torch.int32, device=dev)
         pool_vals = torch.empty(0, dtype=inp.weights.dtype, device=dev)
-        hubs64 = vlists[4].to(torch.int64)
         vlists = vlists[:4] + [vlists[4][:0]]
     else:
         n_pool = int(offsets[-1])
