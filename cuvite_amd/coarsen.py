@@ -48,12 +48,15 @@ def _owner_lookup(comm: Comm, part: Partition, query: torch.Tensor,
     return torch.cat([back[p] for p in range(comm.world)])
 
 
-def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor
+def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor,
+            halo=None
             ) -> Tuple[DistGraph, Callable[[torch.Tensor], torch.Tensor]]:
     """Build the next-level graph from community labels `cvect` (int64 [nv],
     global old-community ids). Returns (new DistGraph for this rank, renum)
     where renum maps any tensor of old community gids to new vertex gids
-    (collective: all ranks must call it together with their own queries)."""
+    (collective: all ranks must call it together with their own queries).
+    `halo`: reuse the phase's HaloContext (ghost set is label-independent,
+    so the rebuild is redundant)."""
     dev = dg.g.device
     base, bound = dg.base, dg.bound
     world, rank = comm.world, comm.rank
@@ -90,7 +93,8 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor
         return out
 
     # --- 2. resolve new ids for all referenced communities -------------------
-    halo = build_halo(dg, comm)
+    if halo is None:
+        halo = build_halo(dg, comm)
     ghost_cvect = exchange_ghost_labels(halo, cvect)
     all_cvect = torch.cat([cvect, ghost_cvect])
     ref_uniq = torch.unique(all_cvect)  # sorted

@@ -239,7 +239,7 @@ def run_phase(dg: DistGraph, comm: Comm, cfg: LouvainConfig,
             break
 
     cvect = state.past_comm
-    return prev_mod, cvect, iters
+    return prev_mod, cvect, iters, state.halo
 
 
 def _one_sweep(state: PhaseState, cfg: LouvainConfig, move_fn,
@@ -412,7 +412,7 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
             times["coloring"] += time.perf_counter() - t0
 
         t0 = time.perf_counter()
-        curr_mod, cvect, iters = run_phase(
+        curr_mod, cvect, iters, phase_halo = run_phase(
             level, comm, cfg, curr_mod, threshold,
             colors=colors, num_colors=num_colors)
         times["clustering"] += time.perf_counter() - t0
@@ -426,14 +426,14 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
             if cfg.one_phase:
                 break
             t0 = time.perf_counter()
-            level, renum = coarsen(level, comm, cvect)
+            level, renum = coarsen(level, comm, cvect, halo=phase_halo)
             # cvect-composed orig_assign holds OLD comm gids; renumber them
             orig_assign = renum(orig_assign)
             times["rebuild"] += time.perf_counter() - t0
         else:
             if cfg.threshold_scaling and not cfg.one_phase and phase < 10:
-                curr_mod2, cvect, iters = run_phase(level, comm, cfg, curr_mod,
-                                                    1.0e-6)
+                curr_mod2, cvect, iters, _ = run_phase(level, comm, cfg,
+                                                       curr_mod, 1.0e-6)
                 tot_iters += iters
                 if (curr_mod2 - curr_mod) > 1.0e-6:
                     orig_assign = remap_labels(level, comm, orig_assign, cvect)
