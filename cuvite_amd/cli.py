@@ -235,6 +235,15 @@ def main(argv=None) -> int:
         print(f"Clustering time: {res.times.get('clustering', 0.0):.3f}s")
         print(f"Rebuild time: {res.times.get('rebuild', 0.0):.3f}s")
         print(f"Total time: {t_total:.3f}s  TEPS: {teps:.4g}")
+        # memory observability (ref getrusage ru_maxrss, main.cpp:142-150)
+        import resource
+        rss_mb = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss / 1024
+        if comm.device.type == "cuda":
+            hbm_gb = torch.cuda.max_memory_allocated() / (1 << 30)
+            print(f"Peak memory: host rss {rss_mb:.0f} MB, "
+                  f"device {hbm_gb:.2f} GB")
+        else:
+            print(f"Peak memory: host rss {rss_mb:.0f} MB")
 
     builtin_truth = getattr(args, "_lfr_truth", None)
     if args.output or args.ground_truth or builtin_truth is not None:
