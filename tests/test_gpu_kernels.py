@@ -232,3 +232,37 @@ def test_hub_class_matches_torch_oracle():
     assert torch.equal(tgt_gpu.cpu().to(torch.int64),
                        tgt_cpu.to(torch.int64))
     assert torch.allclose(cw_gpu.cpu(), cw_cpu)
+
+
+def test_gpu_louvain_variants_match_cpu():
+    """Coloring (-c), ordering (-d) and ET variants on GPU (HIP backend)
+    match the CPU torch-oracle runs on karate exactly (fp64)."""
+    from cuvite_amd.louvain import louvain, LouvainConfig
+    from cuvite_amd.parallel import Comm
+    dev = torch.device("cuda:0")
+    g = karate_graph()
+    for kwargs in ({"coloring": True, "max_colors": 6},
+                   {"ordering": True, "max_colors": 6},
+                   {"early_term": 1},
+                   {"threshold_scaling": True}):
+        cpu = louvain(single_partition(g), Comm(torch.device("cpu")),
+                      LouvainConfig(backend="torch", **kwargs))
+        gpu = louvain(single_partition(g.to(dev)), Comm(dev),
+                      LouvainConfig(backend="hip", **kwargs))
+        assert abs(cpu.modularity - gpu.modularity) < 1e-9, kwargs
+        assert torch.equal(cpu.communities, gpu.communities.cpu()), kwargs
+
+
+def test_gpu_lfr_recovery():
+    """LFR acceptance config on GPU: planted communities recovered."""
+    from cuvite_amd.generators import lfr_graph
+    from cuvite_amd.louvain import louvain, LouvainConfig
+    from cuvite_amd.parallel import Comm
+    from cuvite_amd.compare import compare_communities
+    dev = torch.device("cuda:0")
+    g, truth = lfr_graph(5000, mu=0.4, seed=2)
+    res = louvain(single_partition(g.to(dev)), Comm(dev),
+                  LouvainConfig(backend="hip"))
+    m = compare_communities(truth, res.communities.cpu())
+    assert m["recall"] > 0.75
+    assert m["f_score"] > 0.6
