@@ -144,7 +144,13 @@ def _hub_static(inp, hubs, hdeg):
     if bool(selfmask.any()):
         selfloop.index_add_(0, seg[selfmask], wts[selfmask])
     del selfmask
-    data = (seg, tails_h, wts, selfloop)
+    if os.environ.get("CUVITE_HUB_SEGSORT") and dev.type == "cuda":
+        # int32 copies + per-hub offsets for the rocPRIM segsort path
+        extra = (tails_h.to(torch.int32), seg.to(torch.int32), offs,
+                 wts.to(inp.weights.dtype))
+    else:
+        extra = None
+    data = (seg, tails_h, wts, selfloop, extra)
     if len(_hub_static_cache) > 4:
         _hub_static_cache.clear()
     _hub_static_cache[key] = (inp.rowptr, hubs, data)
@@ -160,24 +166,33 @@ def _hub_moves_sorted_one(inp, hubs, hdeg):
     XCD-L2 footprint (see profiles/ hang bisection)."""
     dev = inp.rowptr.device
     nhub = hubs.numel()
-    seg, tails_h, wts, selfloop = _hub_static(inp, hubs, hdeg)
-    comm = inp.curr_comm[tails_h].to(torch.int64)
+    seg, tails_h, wts, selfloop, extra = _hub_static(inp, hubs, hdeg)
     C = inp.comm_degree.numel()
-    key = seg * C + comm
-    del comm
-    key_s, order = torch.sort(key)
-    del key
-    w_s = wts[order]
-    del wts, order
-    uniq, counts = torch.unique_consecutive(key_s, return_counts=True)
-    del key_s
-    ends = torch.cumsum(counts, dim=0) - 1
-    del counts
-    cs = torch.cumsum(w_s, dim=0)
-    del w_s
-    wsum = cs[ends].clone()
-    wsum[1:] -= cs[ends[:-1]]
-    del cs, ends
+    if extra is not None:
+        # rocPRIM path: segmented narrow-bit radix sort + reduce_by_key
+        tails32, seg32, offs, wts_w = extra
+        uniq, sums, cnt = _require().hub_candidates(
+            tails32, wts_w, seg32, inp.curr_comm, offs, C)
+        ncand = int(cnt[0])
+        uniq = uniq[:ncand]
+        wsum = sums[:ncand].to(torch.float64)
+    else:
+        comm = inp.curr_comm[tails_h].to(torch.int64)
+        key = seg * C + comm
+        del comm
+        key_s, order = torch.sort(key)
+        del key
+        w_s = wts[order]
+        del order
+        uniq, counts = torch.unique_consecutive(key_s, return_counts=True)
+        del key_s
+        ends = torch.cumsum(counts, dim=0) - 1
+        del counts
+        cs = torch.cumsum(w_s, dim=0)
+        del w_s
+        wsum = cs[ends].clone()
+        wsum[1:] -= cs[ends[:-1]]
+        del cs, ends
     hub_of = uniq // C
     y = uniq % C
     del uniq

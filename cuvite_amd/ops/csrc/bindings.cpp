@@ -45,6 +45,16 @@ void launch_csr_place(const int64_t*, const int64_t*, const W*, int64_t,
                       hipStream_t);
 template <typename W>
 void launch_row_sum(const int64_t*, const W*, int64_t, W*, hipStream_t);
+void launch_gather_comm(const int32_t*, const int32_t*, int64_t, int32_t*,
+                        hipStream_t);
+void launch_pack_key64(const int32_t*, const int32_t*, int64_t, int64_t,
+                       int64_t*, hipStream_t);
+template <typename W>
+void segsort_pairs(void*, size_t*, const int32_t*, int32_t*, const W*, W*,
+                   int64_t, int, const int64_t*, int, hipStream_t);
+template <typename W>
+void reduce_by_key64(void*, size_t*, const int64_t*, const W*, int64_t,
+                     int64_t*, W*, unsigned int*, hipStream_t);
 
 }  // namespace cuvite
 
@@ -200,6 +210,71 @@ std::vector<at::Tensor> csr_from_edges(int64_t nv, int64_t base,
   return {rowptr, tails, weights};
 }
 
+std::vector<at::Tensor> hub_candidates(at::Tensor tails_flat,
+                                       at::Tensor weights_flat,
+                                       at::Tensor seg_flat,
+                                       at::Tensor curr_comm, at::Tensor eoffs,
+                                       int64_t C) {
+  CHECK_DEV(tails_flat); CHECK_CONT(tails_flat);
+  CHECK_DEV(weights_flat); CHECK_CONT(weights_flat);
+  CHECK_DEV(seg_flat); CHECK_CONT(seg_flat);
+  CHECK_DEV(curr_comm); CHECK_CONT(curr_comm);
+  CHECK_DEV(eoffs); CHECK_CONT(eoffs);
+  TORCH_CHECK(tails_flat.scalar_type() == at::kInt);
+  TORCH_CHECK(seg_flat.scalar_type() == at::kInt);
+  const int64_t n = tails_flat.numel();
+  const int nseg = (int)(eoffs.numel() - 1);
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  int end_bit = 1;
+  while ((int64_t(1) << end_bit) < C) end_bit++;
+  auto uniq = at::empty({n}, eoffs.options());
+  auto cnt = at::zeros({1}, tails_flat.options());  // int32 count
+  auto sums = at::empty({n}, weights_flat.options());
+  AT_DISPATCH_FLOATING_TYPES(weights_flat.scalar_type(), "hub_cand", [&] {
+    using W = scalar_t;
+    auto keys = at::empty({n}, tails_flat.options());
+    cuvite::launch_gather_comm(tails_flat.data_ptr<int32_t>(),
+                               curr_comm.data_ptr<int32_t>(), n,
+                               keys.data_ptr<int32_t>(), stream);
+    auto keys2 = at::empty({n}, tails_flat.options());
+    auto vals2 = at::empty({n}, weights_flat.options());
+    size_t bytes = 0;
+    cuvite::segsort_pairs<W>(nullptr, &bytes, keys.data_ptr<int32_t>(),
+                             keys2.data_ptr<int32_t>(),
+                             weights_flat.data_ptr<W>(),
+                             vals2.data_ptr<W>(), n, nseg,
+                             eoffs.data_ptr<int64_t>(), end_bit, stream);
+    auto temp = at::empty({(int64_t)bytes},
+                          tails_flat.options().dtype(at::kByte));
+    cuvite::segsort_pairs<W>(temp.data_ptr(), &bytes,
+                             keys.data_ptr<int32_t>(),
+                             keys2.data_ptr<int32_t>(),
+                             weights_flat.data_ptr<W>(),
+                             vals2.data_ptr<W>(), n, nseg,
+                             eoffs.data_ptr<int64_t>(), end_bit, stream);
+    auto key64 = at::empty({n}, eoffs.options());
+    cuvite::launch_pack_key64(keys2.data_ptr<int32_t>(),
+                              seg_flat.data_ptr<int32_t>(), n, C,
+                              key64.data_ptr<int64_t>(), stream);
+    size_t bytes2 = 0;
+    cuvite::reduce_by_key64<W>(nullptr, &bytes2, key64.data_ptr<int64_t>(),
+                               vals2.data_ptr<W>(), n,
+                               uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
+                               (unsigned int*)cnt.data_ptr<int32_t>(),
+                               stream);
+    auto temp2 = at::empty({(int64_t)bytes2},
+                           tails_flat.options().dtype(at::kByte));
+    cuvite::reduce_by_key64<W>(temp2.data_ptr(), &bytes2,
+                               key64.data_ptr<int64_t>(),
+                               vals2.data_ptr<W>(), n,
+                               uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
+                               (unsigned int*)cnt.data_ptr<int32_t>(),
+                               stream);
+  });
+  C10_HIP_CHECK(hipGetLastError());
+  return {uniq, sums, cnt};
+}
+
 at::Tensor row_sum(at::Tensor rowptr, at::Tensor weights) {
   CHECK_DEV(rowptr); CHECK_CONT(rowptr);
   CHECK_DEV(weights); CHECK_CONT(weights);
@@ -227,4 +302,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("csr_from_edges", &csr_from_edges,
         "sort-free CSR assembly on device (HIP)");
   m.def("row_sum", &row_sum, "per-row CSR weight sum (HIP)");
+  m.def("hub_candidates", &hub_candidates,
+        "segmented-sort + reduce_by_key hub candidate generation (rocPRIM)");
 }

@@ -21,6 +21,9 @@
 // match the fp64 CPU oracle (tie-break on equal gains -> smaller GLOBAL id).
 
 #include <hip/hip_runtime.h>
+#include <rocprim/device/device_segmented_radix_sort.hpp>
+#include <rocprim/device/device_reduce_by_key.hpp>
+
 #include <cstdint>
 
 #define DEV_INLINE __device__ __forceinline__
@@ -537,6 +540,33 @@ __global__ __launch_bounds__(BLOCK) void row_sum_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Hub candidate generation via rocPRIM (CUVITE_HUB_SEGSORT experiment):
+// per-hub segmented radix sort of the community keys (only ceil(log2 C) bits)
+// + reduce_by_key for the (hub, community) weight sums. The torch path sorts
+// full 64-bit packed keys; narrow-bit segmented sort does ~half the radix
+// passes.
+// ---------------------------------------------------------------------------
+
+__global__ void gather_comm_kernel(const int32_t* __restrict__ tails_flat,
+                                   const int32_t* __restrict__ curr_comm,
+                                   int64_t n, int32_t* __restrict__ keys) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    keys[i] = curr_comm[tails_flat[i]];
+}
+
+__global__ void pack_key64_kernel(const int32_t* __restrict__ keys,
+                                  const int32_t* __restrict__ seg_flat,
+                                  int64_t n, int64_t C,
+                                  int64_t* __restrict__ key64) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    key64[i] = (int64_t)seg_flat[i] * C + keys[i];
+}
+
 // ------------------------------- launchers ---------------------------------
 
 static int grid_for(int64_t n, int block) {
@@ -694,6 +724,21 @@ template void launch_csr_place<double>(const int64_t*, const int64_t*,
                                        const int64_t*, int32_t*, int64_t*,
                                        double*, hipStream_t);
 
+void launch_gather_comm(const int32_t* tails_flat, const int32_t* curr_comm,
+                        int64_t n, int32_t* keys, hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL(gather_comm_kernel, dim3(grid_for(n, 256)), dim3(256), 0,
+                     stream, tails_flat, curr_comm, n, keys);
+}
+
+void launch_pack_key64(const int32_t* keys, const int32_t* seg_flat,
+                       int64_t n, int64_t C, int64_t* key64,
+                       hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL(pack_key64_kernel, dim3(grid_for(n, 256)), dim3(256), 0,
+                     stream, keys, seg_flat, n, C, key64);
+}
+
 template <typename W>
 void launch_row_sum(const int64_t* rowptr, const W* weights, int64_t nv,
                     W* out, hipStream_t stream) {
@@ -709,5 +754,37 @@ template void launch_row_sum<float>(const int64_t*, const float*, int64_t,
                                     float*, hipStream_t);
 template void launch_row_sum<double>(const int64_t*, const double*, int64_t,
                                      double*, hipStream_t);
+
+// rocPRIM wrappers (two-phase: bytes query with temp==nullptr, then run).
+template <typename W>
+void segsort_pairs(void* temp, size_t* bytes, const int32_t* keys_in,
+                   int32_t* keys_out, const W* vals_in, W* vals_out,
+                   int64_t n, int nseg, const int64_t* offs, int end_bit,
+                   hipStream_t stream) {
+  (void)rocprim::segmented_radix_sort_pairs(
+      temp, *bytes, keys_in, keys_out, vals_in, vals_out, (size_t)n,
+      (unsigned)nseg, offs, offs + 1, 0u, (unsigned)end_bit, stream);
+}
+template void segsort_pairs<float>(void*, size_t*, const int32_t*, int32_t*,
+                                   const float*, float*, int64_t, int,
+                                   const int64_t*, int, hipStream_t);
+template void segsort_pairs<double>(void*, size_t*, const int32_t*, int32_t*,
+                                    const double*, double*, int64_t, int,
+                                    const int64_t*, int, hipStream_t);
+
+template <typename W>
+void reduce_by_key64(void* temp, size_t* bytes, const int64_t* keys,
+                     const W* vals, int64_t n, int64_t* uniq_out, W* sums_out,
+                     unsigned int* count_out, hipStream_t stream) {
+  (void)rocprim::reduce_by_key(temp, *bytes, keys, vals, (size_t)n, uniq_out,
+                               sums_out, count_out, rocprim::plus<W>(),
+                               rocprim::equal_to<int64_t>(), stream);
+}
+template void reduce_by_key64<float>(void*, size_t*, const int64_t*,
+                                     const float*, int64_t, int64_t*, float*,
+                                     unsigned int*, hipStream_t);
+template void reduce_by_key64<double>(void*, size_t*, const int64_t*,
+                                      const double*, int64_t, int64_t*,
+                                      double*, unsigned int*, hipStream_t);
 
 }  // namespace cuvite

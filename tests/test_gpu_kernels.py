@@ -2,6 +2,8 @@
 against the plain PyTorch fp32/fp64 reference of the same op
 (cuvite_amd.local_move.local_move_torch)."""
 
+import os
+
 import pytest
 import torch
 
@@ -266,3 +268,35 @@ def test_gpu_lfr_recovery():
     m = compare_communities(truth, res.communities.cpu())
     assert m["recall"] > 0.75
     assert m["f_score"] > 0.6
+
+
+@pytest.mark.skipif(not os.environ.get("CUVITE_TEST_SEGSORT"),
+                    reason="experimental rocPRIM segsort hub path "
+                           "(set CUVITE_TEST_SEGSORT=1)")
+def test_hub_segsort_matches_default():
+    """CUVITE_HUB_SEGSORT=1 (rocPRIM segmented sort + reduce_by_key) must
+    match the default torch-sort hub path exactly on unit weights."""
+    from cuvite_amd import ops
+    torch.manual_seed(11)
+    nv = 8192
+    src = torch.cat([torch.zeros(40000, dtype=torch.int64),
+                     torch.randint(1, nv, (nv * 8,))])
+    dst = torch.cat([torch.randint(0, nv, (40000,)),
+                     torch.randint(0, nv, (nv * 8,))])
+    w = torch.ones(src.numel(), dtype=torch.float64)
+    g = Graph.from_edge_tuples(nv, src, dst, w)
+    dev = torch.device("cuda:0")
+    inp = _inputs(g, dev, "random", seed=5)
+    ops._bucket_cache.clear()
+    ops._hub_static_cache.clear()
+    t_ref, cw_ref = ops.local_move(inp)
+    os.environ["CUVITE_HUB_SEGSORT"] = "1"
+    try:
+        ops._bucket_cache.clear()
+        ops._hub_static_cache.clear()
+        t_new, cw_new = ops.local_move(inp)
+    finally:
+        del os.environ["CUVITE_HUB_SEGSORT"]
+        ops._hub_static_cache.clear()
+    assert torch.equal(t_ref, t_new)
+    assert torch.allclose(cw_ref, cw_new)
