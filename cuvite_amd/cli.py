@@ -198,7 +198,7 @@ def main(argv=None) -> int:
         print(f"Graph: nv={dg.nv_global} ne(directed)={int(ne_global)} "
               f"ranks={comm.world} device={comm.device.type} "
               f"ingest={t_ingest:.3f}s")
-    if args.stats:
+    if args.stats or args.just_process:
         print_dist_stats(dg, comm)
     if args.just_process:
         return 0
