@@ -147,7 +147,7 @@ def _hub_static(inp, hubs, hdeg):
     if os.environ.get("CUVITE_HUB_SEGSORT") and dev.type == "cuda":
         # int32 copies + per-hub offsets for the rocPRIM segsort path
         extra = (tails_h.to(torch.int32), seg.to(torch.int32), offs,
-                 wts.to(inp.weights.dtype))
+                 wts.to(inp.weights.dtype), hubs.to(torch.int32))
     else:
         extra = None
     data = (seg, tails_h, wts, selfloop, extra)
@@ -169,30 +169,30 @@ def _hub_moves_sorted_one(inp, hubs, hdeg):
     seg, tails_h, wts, selfloop, extra = _hub_static(inp, hubs, hdeg)
     C = inp.comm_degree.numel()
     if extra is not None:
-        # rocPRIM path: segmented narrow-bit radix sort + reduce_by_key
-        tails32, seg32, offs, wts_w = extra
-        uniq, sums, cnt = _require().hub_candidates(
-            tails32, wts_w, seg32, inp.curr_comm, offs, C)
-        ncand = int(cnt[0])
-        uniq = uniq[:ncand]
-        wsum = sums[:ncand].to(torch.float64)
-    else:
-        comm = inp.curr_comm[tails_h].to(torch.int64)
-        key = seg * C + comm
-        del comm
-        key_s, order = torch.sort(key)
-        del key
-        w_s = wts[order]
-        del order
-        uniq, counts = torch.unique_consecutive(key_s, return_counts=True)
-        del key_s
-        ends = torch.cumsum(counts, dim=0) - 1
-        del counts
-        cs = torch.cumsum(w_s, dim=0)
-        del w_s
-        wsum = cs[ends].clone()
-        wsum[1:] -= cs[ends[:-1]]
-        del cs, ends
+        # rocPRIM path, fully device-side: segmented narrow-bit radix sort
+        # + reduce_by_key + wave-per-hub argmax (no host sync per iteration)
+        tails32, seg32, offs, wts_w, hubs32 = extra
+        tgt_hub, cw_hub = _require().hub_moves(
+            tails32, wts_w, seg32, offs, hubs32, selfloop,
+            inp.curr_comm, inp.v_degree, inp.comm_size, inp.comm_degree,
+            inp.comm_gid, float(inp.constant))
+        return tgt_hub, cw_hub
+    comm = inp.curr_comm[tails_h].to(torch.int64)
+    key = seg * C + comm
+    del comm
+    key_s, order = torch.sort(key)
+    del key
+    w_s = wts[order]
+    del order
+    uniq, counts = torch.unique_consecutive(key_s, return_counts=True)
+    del key_s
+    ends = torch.cumsum(counts, dim=0) - 1
+    del counts
+    cs = torch.cumsum(w_s, dim=0)
+    del w_s
+    wsum = cs[ends].clone()
+    wsum[1:] -= cs[ends[:-1]]
+    del cs, ends
     hub_of = uniq // C
     y = uniq % C
     del uniq

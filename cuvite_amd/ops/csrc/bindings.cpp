@@ -55,6 +55,11 @@ void segsort_pairs(void*, size_t*, const int32_t*, int32_t*, const W*, W*,
 template <typename W>
 void reduce_by_key64(void*, size_t*, const int64_t*, const W*, int64_t,
                      int64_t*, W*, unsigned int*, hipStream_t);
+template <typename W>
+void launch_hub_argmax(const int64_t*, const W*, const int32_t*, int64_t,
+                       const int32_t*, int, const double*, const int32_t*,
+                       const W*, const int64_t*, const W*, const int64_t*,
+                       double, int32_t*, W*, hipStream_t);
 
 }  // namespace cuvite
 
@@ -275,6 +280,80 @@ std::vector<at::Tensor> hub_candidates(at::Tensor tails_flat,
   return {uniq, sums, cnt};
 }
 
+// Full hub move: gather -> segmented sort -> reduce_by_key -> wave-per-hub
+// argmax, all device-side (no host sync). Returns per-hub (target, wcc).
+std::vector<at::Tensor> hub_moves(
+    at::Tensor tails_flat, at::Tensor weights_flat, at::Tensor seg_flat,
+    at::Tensor eoffs, at::Tensor hubs_i32, at::Tensor hub_self,
+    at::Tensor curr_comm, at::Tensor v_degree, at::Tensor comm_size,
+    at::Tensor comm_degree, at::Tensor comm_gid, double constant) {
+  CHECK_DEV(tails_flat); CHECK_CONT(tails_flat);
+  CHECK_DEV(hubs_i32); CHECK_CONT(hubs_i32);
+  TORCH_CHECK(hubs_i32.scalar_type() == at::kInt);
+  TORCH_CHECK(hub_self.scalar_type() == at::kDouble);
+  const int64_t n = tails_flat.numel();
+  const int nhub = (int)hubs_i32.numel();
+  const int64_t C = comm_degree.numel();
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  int end_bit = 1;
+  while ((int64_t(1) << end_bit) < C) end_bit++;
+  auto target_hub = at::empty({nhub}, hubs_i32.options());
+  auto cw_hub = at::empty({nhub}, weights_flat.options());
+  AT_DISPATCH_FLOATING_TYPES(weights_flat.scalar_type(), "hub_moves", [&] {
+    using W = scalar_t;
+    auto keys = at::empty({n}, tails_flat.options());
+    cuvite::launch_gather_comm(tails_flat.data_ptr<int32_t>(),
+                               curr_comm.data_ptr<int32_t>(), n,
+                               keys.data_ptr<int32_t>(), stream);
+    auto keys2 = at::empty({n}, tails_flat.options());
+    auto vals2 = at::empty({n}, weights_flat.options());
+    size_t bytes = 0;
+    cuvite::segsort_pairs<W>(nullptr, &bytes, keys.data_ptr<int32_t>(),
+                             keys2.data_ptr<int32_t>(),
+                             weights_flat.data_ptr<W>(), vals2.data_ptr<W>(),
+                             n, nhub, eoffs.data_ptr<int64_t>(), end_bit,
+                             stream);
+    auto temp = at::empty({(int64_t)bytes},
+                          tails_flat.options().dtype(at::kByte));
+    cuvite::segsort_pairs<W>(temp.data_ptr(), &bytes,
+                             keys.data_ptr<int32_t>(),
+                             keys2.data_ptr<int32_t>(),
+                             weights_flat.data_ptr<W>(), vals2.data_ptr<W>(),
+                             n, nhub, eoffs.data_ptr<int64_t>(), end_bit,
+                             stream);
+    auto key64 = at::empty({n}, eoffs.options());
+    cuvite::launch_pack_key64(keys2.data_ptr<int32_t>(),
+                              seg_flat.data_ptr<int32_t>(), n, C,
+                              key64.data_ptr<int64_t>(), stream);
+    auto uniq = at::empty({n}, eoffs.options());
+    auto sums = at::empty({n}, weights_flat.options());
+    auto cnt = at::zeros({1}, tails_flat.options());
+    size_t bytes2 = 0;
+    cuvite::reduce_by_key64<W>(nullptr, &bytes2, key64.data_ptr<int64_t>(),
+                               vals2.data_ptr<W>(), n,
+                               uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
+                               (unsigned int*)cnt.data_ptr<int32_t>(),
+                               stream);
+    auto temp2 = at::empty({(int64_t)bytes2},
+                           tails_flat.options().dtype(at::kByte));
+    cuvite::reduce_by_key64<W>(temp2.data_ptr(), &bytes2,
+                               key64.data_ptr<int64_t>(),
+                               vals2.data_ptr<W>(), n,
+                               uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
+                               (unsigned int*)cnt.data_ptr<int32_t>(),
+                               stream);
+    cuvite::launch_hub_argmax<W>(
+        uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
+        cnt.data_ptr<int32_t>(), C, hubs_i32.data_ptr<int32_t>(), nhub,
+        hub_self.data_ptr<double>(), curr_comm.data_ptr<int32_t>(),
+        v_degree.data_ptr<W>(), comm_size.data_ptr<int64_t>(),
+        comm_degree.data_ptr<W>(), comm_gid.data_ptr<int64_t>(), constant,
+        target_hub.data_ptr<int32_t>(), cw_hub.data_ptr<W>(), stream);
+  });
+  C10_HIP_CHECK(hipGetLastError());
+  return {target_hub, cw_hub};
+}
+
 at::Tensor row_sum(at::Tensor rowptr, at::Tensor weights) {
   CHECK_DEV(rowptr); CHECK_CONT(rowptr);
   CHECK_DEV(weights); CHECK_CONT(weights);
@@ -304,4 +383,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_sum", &row_sum, "per-row CSR weight sum (HIP)");
   m.def("hub_candidates", &hub_candidates,
         "segmented-sort + reduce_by_key hub candidate generation (rocPRIM)");
+  m.def("hub_moves", &hub_moves,
+        "full device-side hub move: segsort + reduce_by_key + argmax");
 }

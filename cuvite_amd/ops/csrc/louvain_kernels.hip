@@ -567,6 +567,66 @@ __global__ void pack_key64_kernel(const int32_t* __restrict__ keys,
     key64[i] = (int64_t)seg_flat[i] * C + keys[i];
 }
 
+// One wave per hub over its (sorted) candidate range: wcc pass, then the
+// exact-tie-break dQ argmax (same Best semantics as the class kernels).
+template <typename W, int BLOCK>
+__global__ __launch_bounds__(BLOCK) void hub_argmax_kernel(
+    const int64_t* __restrict__ uniq, const W* __restrict__ sums,
+    const int32_t* __restrict__ cnt_ptr, int64_t C,
+    const int32_t* __restrict__ hubs, int nhub,
+    const double* __restrict__ hub_self,
+    const int32_t* __restrict__ curr_comm, const W* __restrict__ v_degree,
+    const int64_t* __restrict__ comm_size, const W* __restrict__ comm_degree,
+    const int64_t* __restrict__ comm_gid, double constant,
+    int32_t* __restrict__ target_hub, W* __restrict__ cw_hub) {
+  constexpr int WAVES = BLOCK / 64;
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const int hidx = blockIdx.x * WAVES + wave;
+  if (hidx >= nhub) return;
+  const int64_t n = (int64_t)cnt_ptr[0];
+  const int32_t v = hubs[hidx];
+  const int32_t cc = curr_comm[v];
+  const int64_t keylo = (int64_t)hidx * C;
+  const int64_t keyhi = keylo + C;
+  int64_t lo = 0, hi = n;
+  while (lo < hi) {
+    const int64_t m = (lo + hi) >> 1;
+    if (uniq[m] < keylo) lo = m + 1; else hi = m;
+  }
+  const int64_t r0 = lo;
+  hi = n;
+  while (lo < hi) {
+    const int64_t m = (lo + hi) >> 1;
+    if (uniq[m] < keyhi) lo = m + 1; else hi = m;
+  }
+  const int64_t r1 = lo;
+  double wcc = 0.0;
+  for (int64_t i = r0 + lane; i < r1; i += 64)
+    if ((int32_t)(uniq[i] - keylo) == cc) wcc += (double)sums[i];
+  wcc = sum_reduce<64>(wcc);
+  wcc = __shfl(wcc, 0, 64);
+  const double eix = wcc - hub_self[hidx];
+  const double vdeg = (double)v_degree[v];
+  const double ax = (double)comm_degree[cc] - vdeg;
+  Best best{0.0, comm_gid[cc], cc};
+  for (int64_t i = r0 + lane; i < r1; i += 64) {
+    const int32_t y = (int32_t)(uniq[i] - keylo);
+    if (y == cc) continue;
+    const double eiy = (double)sums[i];
+    const double ay = (double)comm_degree[y];
+    const double g = 2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant;
+    best_combine(best, g, comm_gid[y], y);
+  }
+  best_reduce<64>(best);
+  if (lane == 0) {
+    int32_t tgt = best.dense;
+    if (comm_size[tgt] == 1 && comm_size[cc] == 1 && best.gid > comm_gid[cc])
+      tgt = cc;
+    target_hub[hidx] = tgt;
+    cw_hub[hidx] = (W)wcc;
+  }
+}
+
 // ------------------------------- launchers ---------------------------------
 
 static int grid_for(int64_t n, int block) {
@@ -738,6 +798,38 @@ void launch_pack_key64(const int32_t* keys, const int32_t* seg_flat,
   hipLaunchKernelGGL(pack_key64_kernel, dim3(grid_for(n, 256)), dim3(256), 0,
                      stream, keys, seg_flat, n, C, key64);
 }
+
+template <typename W>
+void launch_hub_argmax(const int64_t* uniq, const W* sums,
+                       const int32_t* cnt, int64_t C, const int32_t* hubs,
+                       int nhub, const double* hub_self,
+                       const int32_t* curr_comm, const W* v_degree,
+                       const int64_t* comm_size, const W* comm_degree,
+                       const int64_t* comm_gid, double constant,
+                       int32_t* target_hub, W* cw_hub, hipStream_t stream) {
+  if (nhub == 0) return;
+  constexpr int BLOCK = 256;
+  constexpr int WAVES = BLOCK / 64;
+  hipLaunchKernelGGL((hub_argmax_kernel<W, BLOCK>),
+                     dim3((nhub + WAVES - 1) / WAVES), dim3(BLOCK), 0, stream,
+                     uniq, sums, cnt, C, hubs, nhub, hub_self, curr_comm,
+                     v_degree, comm_size, comm_degree, comm_gid, constant,
+                     target_hub, cw_hub);
+}
+template void launch_hub_argmax<float>(const int64_t*, const float*,
+                                       const int32_t*, int64_t,
+                                       const int32_t*, int, const double*,
+                                       const int32_t*, const float*,
+                                       const int64_t*, const float*,
+                                       const int64_t*, double, int32_t*,
+                                       float*, hipStream_t);
+template void launch_hub_argmax<double>(const int64_t*, const double*,
+                                        const int32_t*, int64_t,
+                                        const int32_t*, int, const double*,
+                                        const int32_t*, const double*,
+                                        const int64_t*, const double*,
+                                        const int64_t*, double, int32_t*,
+                                        double*, hipStream_t);
 
 template <typename W>
 void launch_row_sum(const int64_t* rowptr, const W* weights, int64_t nv,
