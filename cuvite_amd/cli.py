@@ -95,6 +95,9 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--diag-files", action="store_true",
                     help="write per-rank diagnostics to dat.out.<rank> "
                     "instead of stdout (ref main.cpp:101-110)")
+    ap.add_argument("--unit-weights", action="store_true",
+                    help="force all edge weights to 1.0 for file inputs "
+                    "(ref SET_EDGE_WEIGHTS_TO_ONE, distgraph.cpp:200-202)")
     ap.add_argument("--stats", action="store_true",
                     help="print the graph distribution table "
                     "(ref printStats, distgraph.hpp:100-149)")
@@ -128,6 +131,7 @@ def _ingest(args, comm: Comm) -> DistGraph:
     if args.input:
         return load_dist_graph(args.input, comm.rank, comm.world,
                                balanced=args.balanced,
+                               unit_weights=args.unit_weights,
                                weight_dtype=wdtype).to(dev)
     if args.karate:
         g = karate_graph(wdtype)

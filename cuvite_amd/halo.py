@@ -38,6 +38,7 @@ class HaloContext:
     recv_offsets: torch.Tensor    # int64 [world+1]: ghosts grouped by owner rank
     send_buf: List[torch.Tensor] = field(default_factory=list)
     recv_buf: List[torch.Tensor] = field(default_factory=list)
+    wire_dtype: torch.dtype = torch.int64
 
     @property
     def ng(self) -> int:
@@ -89,10 +90,15 @@ def build_halo(dg: DistGraph, comm: Comm) -> HaloContext:
             send_idx.append(got[p] - base)
 
     ctx = HaloContext(dg, comm, ghosts, tails_dense, send_idx, recv_offsets)
-    ctx.send_buf = [torch.empty(s.numel(), dtype=torch.int64, device=dev)
+    # community labels are vertex gids; nv_global < 2^31 (every BASELINE
+    # config up to R-MAT s30) lets the per-iteration label exchange ride an
+    # int32 wire format - half the xGMI bytes
+    wire = torch.int32 if dg.nv_global < (1 << 31) else torch.int64
+    ctx.wire_dtype = wire
+    ctx.send_buf = [torch.empty(s.numel(), dtype=wire, device=dev)
                     for s in send_idx]
     ctx.recv_buf = [torch.empty(int(recv_offsets[p + 1] - recv_offsets[p]),
-                                dtype=torch.int64, device=dev)
+                                dtype=wire, device=dev)
                     for p in range(world)]
     return ctx
 
@@ -106,7 +112,7 @@ def exchange_ghost_labels(ctx: HaloContext, curr_comm: torch.Tensor) -> torch.Te
         return torch.empty(0, dtype=torch.int64, device=curr_comm.device)
     for p in range(ctx.comm.world):
         if ctx.send_idx[p].numel():
-            torch.index_select(curr_comm, 0, ctx.send_idx[p], out=ctx.send_buf[p])
+            ctx.send_buf[p].copy_(curr_comm[ctx.send_idx[p]])
     ctx.comm.exchange_fixed(ctx.send_buf, ctx.recv_buf)
     out = torch.empty(ctx.ng, dtype=torch.int64, device=curr_comm.device)
     for p in range(ctx.comm.world):
