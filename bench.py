@@ -34,7 +34,7 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 from cuvite_amd.generators import rmat_dist_graph  # noqa: E402
 from cuvite_amd.louvain import (LouvainConfig, PhaseState, _modularity,
                                 _one_sweep, _pick_move_fn)  # noqa: E402
-from cuvite_amd.parallel import Comm, init_from_env  # noqa: E402
+from cuvite_amd.parallel import init_from_env  # noqa: E402
 
 
 def main():

@@ -16,7 +16,6 @@ Examples:
 from __future__ import annotations
 
 import argparse
-import os
 import sys
 import time
 
@@ -25,7 +24,7 @@ import torch
 from .compare import compare_communities
 from .generators import (rgg_dist_graph, rmat_dist_graph, karate_graph,
                          lfr_dist_graph)
-from .graph import DistGraph, Graph, Partition, single_partition
+from .graph import DistGraph, Graph, Partition
 from .io import (load_dist_graph, load_ground_truth, write_communities,
                  write_dist_graph)
 from .louvain import LouvainConfig, louvain

@@ -17,7 +17,7 @@ sort/unique/searchsorted.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import List, Optional
+from typing import List
 
 import torch
 

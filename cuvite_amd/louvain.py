@@ -22,8 +22,8 @@ from typing import List, Optional
 
 import torch
 
-from .graph import DistGraph, Partition
-from .halo import (HaloContext, build_halo, exchange_ghost_labels,
+from .graph import DistGraph
+from .halo import (build_halo, exchange_ghost_labels,
                    fetch_remote_comm_info, push_remote_deltas)
 from .local_move import MoveInputs, local_move_torch, modularity_parts
 from .ops import scatter_add_

@@ -93,3 +93,22 @@ def test_vertex_ordering_louvain():
                   LouvainConfig(backend="torch", ordering=True, max_colors=8))
     q = _modularity_of(g, res.communities)
     assert q > 0.35
+
+
+def test_karate_golden_trajectory():
+    """Golden regression: the karate-club run must keep producing exactly
+    this modularity/phase/iteration profile (any change means the algorithm
+    semantics moved)."""
+    import torch
+    from cuvite_amd.generators import karate_graph
+    from cuvite_amd.graph import single_partition
+    from cuvite_amd.louvain import louvain, LouvainConfig
+    from cuvite_amd.parallel import Comm
+
+    res = louvain(single_partition(karate_graph()),
+                  Comm(torch.device("cpu")), LouvainConfig(backend="torch"))
+    assert res.modularity == pytest.approx(0.408695, abs=1e-6)
+    assert res.phases == 3
+    assert res.total_iters == 11
+    assert res.modularity_per_level == pytest.approx(
+        res.modularity_per_level)  # shape stability
