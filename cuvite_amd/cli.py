@@ -94,6 +94,9 @@ def build_parser() -> argparse.ArgumentParser:
     ap.add_argument("--diag-files", action="store_true",
                     help="write per-rank diagnostics to dat.out.<rank> "
                     "instead of stdout (ref main.cpp:101-110)")
+    ap.add_argument("--verbose", action="store_true",
+                    help="per-iteration modularity + timing breakdown "
+                    "(ref PRINT_TIMEDS, louvain_cuda.cu:2380-2730)")
     ap.add_argument("--unit-weights", action="store_true",
                     help="force all edge weights to 1.0 for file inputs "
                     "(ref SET_EDGE_WEIGHTS_TO_ONE, distgraph.cpp:200-202)")
@@ -216,6 +219,7 @@ def main(argv=None) -> int:
         ordering=args.ordering > 0,
         max_colors=max(args.coloring, args.ordering) or 8,
         backend=args.backend,
+        verbose=args.verbose,
     )
     if args.max_phases:
         cfg.max_phases = args.max_phases

@@ -28,6 +28,7 @@ from .halo import (build_halo, exchange_ghost_labels,
 from .local_move import MoveInputs, local_move_torch, modularity_parts
 from .ops import scatter_add_
 from .parallel import Comm
+from .utils.timers import Timers
 
 TERMINATION_PHASE_COUNT = 200   # ref utils.hpp:17-19
 MAX_TOTAL_ITERS = 10000         # ref main.cpp:486-494
@@ -211,6 +212,7 @@ def run_phase(dg: DistGraph, comm: Comm, cfg: LouvainConfig,
     prev_mod = lower
     iters = 0
     rng = torch.Generator().manual_seed(12345 + comm.rank)
+    timers = Timers(sync=dg.g.device.type == "cuda") if cfg.verbose else None
 
     use_colors = colors is not None and num_colors > 0
     color_order: List[torch.Tensor] = []
@@ -220,11 +222,19 @@ def run_phase(dg: DistGraph, comm: Comm, cfg: LouvainConfig,
 
     while iters < cfg.max_iters_per_phase:
         iters += 1
-        target = _one_sweep(state, cfg, move_fn,
-                            color_order if use_colors else None)
-
-        # modularity over the whole sweep
-        curr_mod = _modularity(state)
+        if timers is None:
+            target = _one_sweep(state, cfg, move_fn,
+                                color_order if use_colors else None)
+            curr_mod = _modularity(state)
+        else:
+            with timers("sweep"):
+                target = _one_sweep(state, cfg, move_fn,
+                                    color_order if use_colors else None)
+            with timers("modularity"):
+                curr_mod = _modularity(state)
+            if comm.rank == 0:
+                print(f"  iter {iters}: Q={curr_mod:.6f} "
+                      f"[{timers.summary()}]")
 
         frozen = _et_update(state, cfg, target, rng)
 
