@@ -22,12 +22,12 @@ import os
 import sys
 import time
 
+import torch
+
 
 def _p(msg):
     print(f"[bench {time.strftime('%H:%M:%S')}] {msg}", file=sys.stderr,
           flush=True)
-
-import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
