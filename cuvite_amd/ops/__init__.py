@@ -34,7 +34,7 @@ def _require():
 
 
 # Degree-class boundaries (see louvain_kernels.hip header comment).
-_CLASS_BOUNDS = (16, 64, 512, 4096)
+_CLASS_BOUNDS = (16, 64, 512, 2048, 4096)
 _bucket_cache: dict = {}
 
 
@@ -46,15 +46,16 @@ def _buckets_for(rowptr: torch.Tensor):
     if hit is not None and hit[0] is rowptr:  # identity check: ptr reuse safe
         return hit[1], hit[2], hit[3], hit[4], hit[5]
     deg = rowptr[1:] - rowptr[:-1]
-    b0, b1, b2, b3 = _CLASS_BOUNDS
+    b0, b1, b2, b3, b4 = _CLASS_BOUNDS
     vlists = [
         ((deg > 0) & (deg <= b0)).nonzero(as_tuple=True)[0].to(torch.int32),
         ((deg > b0) & (deg <= b1)).nonzero(as_tuple=True)[0].to(torch.int32),
         ((deg > b1) & (deg <= b2)).nonzero(as_tuple=True)[0].to(torch.int32),
         ((deg > b2) & (deg <= b3)).nonzero(as_tuple=True)[0].to(torch.int32),
-        (deg > b3).nonzero(as_tuple=True)[0].to(torch.int32),
+        ((deg > b3) & (deg <= b4)).nonzero(as_tuple=True)[0].to(torch.int32),
+        (deg > b4).nonzero(as_tuple=True)[0].to(torch.int32),
     ]
-    hubs = vlists[4]
+    hubs = vlists[5]
     if hubs.numel():
         hdeg = deg[hubs.to(torch.int64)]
         caps = torch.pow(
@@ -241,7 +242,7 @@ def local_move(inp):
         n_pool = 0
         pool_keys = torch.empty(0, dtype=torch.int32, device=dev)
         pool_vals = torch.empty(0, dtype=inp.weights.dtype, device=dev)
-        vlists = vlists[:4] + [vlists[4][:0]]
+        vlists = vlists[:5] + [vlists[5][:0]]
     else:
         n_pool = int(offsets[-1])
         # hub-table pool is phase-static: allocate once, reset per iteration
@@ -267,14 +268,14 @@ def local_move(inp):
         print(f"[move] classes {sizes} pool={n_pool}", file=sys.stderr,
               flush=True)
         outs = []
-        for i in range(5):
+        for i in range(6):
             one = [v if j == i else v[:0] for j, v in enumerate(vlists)]
             t0 = time.perf_counter()
             outs.append(ext.local_move_bucketed(
                 inp.rowptr, inp.tails, inp.weights, inp.curr_comm,
                 inp.v_degree, inp.comm_size, inp.comm_degree, inp.comm_gid,
                 float(inp.constant), one, offsets, eoffs,
-                n_hub_edges if i == 4 else 0, pool_keys, pool_vals))
+                n_hub_edges if i == 5 else 0, pool_keys, pool_vals))
             torch.cuda.synchronize()
             print(f"[move] class {i} n={sizes[i]} "
                   f"{time.perf_counter() - t0:.3f}s", file=sys.stderr,
@@ -282,7 +283,7 @@ def local_move(inp):
         # merge: each class wrote its own vertices; take per-class targets
         target = outs[0][0]
         cw = outs[0][1]
-        for i in range(1, 5):
+        for i in range(1, 6):
             vl = vlists[i].to(torch.int64)
             if vl.numel():
                 target[vl] = outs[i][0][vl]

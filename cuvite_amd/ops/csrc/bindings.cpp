@@ -26,7 +26,7 @@ struct MoveArgs {
 
 template <typename W, int LANES, int CAP>
 void launch_sub(const int32_t*, int, const MoveArgs<W>&, hipStream_t);
-template <typename W>
+template <typename W, int CAP>
 void launch_block(const int32_t*, int, const MoveArgs<W>&, hipStream_t);
 template <typename W>
 void launch_hub(const int32_t*, int, const int64_t*, int64_t, const int64_t*,
@@ -102,7 +102,7 @@ std::vector<at::Tensor> local_move(
   CHECK_DEV(curr_comm); CHECK_CONT(curr_comm);
   TORCH_CHECK(tails.scalar_type() == at::kInt, "tails must be int32");
   TORCH_CHECK(curr_comm.scalar_type() == at::kInt, "curr_comm must be int32");
-  TORCH_CHECK(vlists.size() == 5, "expected 5 degree-class vertex lists");
+  TORCH_CHECK(vlists.size() == 6, "expected 6 degree-class vertex lists");
 
   const int64_t nv = rowptr.numel() - 1;
   auto target = curr_comm.narrow(0, 0, nv).clone();
@@ -124,10 +124,13 @@ std::vector<at::Tensor> local_move(
       cuvite::launch_sub<W, 64, 1024>(vlists[2].data_ptr<int32_t>(),
                                       (int)vlists[2].numel(), args, stream);
     if (vlists[3].numel())
-      cuvite::launch_block<W>(vlists[3].data_ptr<int32_t>(),
-                              (int)vlists[3].numel(), args, stream);
-    if (vlists[4].numel()) {
-      const int nhub = (int)vlists[4].numel();
+      cuvite::launch_block<W, 4096>(vlists[3].data_ptr<int32_t>(),
+                                    (int)vlists[3].numel(), args, stream);
+    if (vlists[4].numel())
+      cuvite::launch_block<W, 8192>(vlists[4].data_ptr<int32_t>(),
+                                    (int)vlists[4].numel(), args, stream);
+    if (vlists[5].numel()) {
+      const int nhub = (int)vlists[5].numel();
       auto hub_self = at::zeros({nhub}, rowptr.options().dtype(at::kDouble));
       auto overflow = at::zeros({1}, rowptr.options().dtype(at::kInt));
       auto p_gain = at::empty({nhub * cuvite::HUB_SPLITS_HOST},
@@ -136,7 +139,7 @@ std::vector<at::Tensor> local_move(
                              rowptr.options().dtype(at::kLong));
       auto p_dense = at::empty({nhub * cuvite::HUB_SPLITS_HOST},
                                rowptr.options().dtype(at::kInt));
-      cuvite::launch_hub<W>(vlists[4].data_ptr<int32_t>(), nhub,
+      cuvite::launch_hub<W>(vlists[5].data_ptr<int32_t>(), nhub,
                             hub_eoffs.data_ptr<int64_t>(), total_hub_edges,
                             global_offsets.data_ptr<int64_t>(),
                             pool_keys.data_ptr<int32_t>(),

@@ -13,7 +13,10 @@
 //   class 0: deg in [1, 16]      16 lanes/vertex, 32-slot LDS table
 //   class 1: deg in (16, 64]     one wave/vertex, 128-slot LDS table
 //   class 2: deg in (64, 512]    one wave/vertex, 1024-slot LDS table
-//   block  : deg in (512, 4096]  one 256-thread block/vertex, 8192-slot LDS
+//   block A: deg in (512, 2048]  one 256-thread block/vertex, 4096-slot LDS
+//                                (48 KB -> 3 blocks/CU; the 8192-slot table
+//                                is 96 KB and caps residency at 1 block/CU)
+//   block B: deg in (2048, 4096] one 256-thread block/vertex, 8192-slot LDS
 //   hub    : deg > 4096          edge-parallel 3-kernel pipeline over a
 //                                global-memory table pool (see below)
 //
@@ -669,11 +672,10 @@ void launch_sub(const int32_t* vlist, int nlist, const MoveArgs<W>& a,
                      a.constant, a.target, a.cluster_weight);
 }
 
-template <typename W>
+template <typename W, int CAP>
 void launch_block(const int32_t* vlist, int nlist, const MoveArgs<W>& a,
                   hipStream_t stream) {
   constexpr int BLOCK = 256;
-  constexpr int CAP = 8192;
   const size_t shmem = (size_t)CAP * (sizeof(W) + sizeof(int32_t));
   auto kern = lv_move_block<W, CAP, BLOCK>;
   if (shmem > 65536)
@@ -720,8 +722,10 @@ void launch_hub(const int32_t* vlist, int nlist, const int64_t* eoffs,
                                        const MoveArgs<W>&, hipStream_t);     \
   template void launch_sub<W, 64, 1024>(const int32_t*, int,                 \
                                         const MoveArgs<W>&, hipStream_t);    \
-  template void launch_block<W>(const int32_t*, int, const MoveArgs<W>&,     \
-                                hipStream_t);                                 \
+  template void launch_block<W, 4096>(const int32_t*, int,                  \
+                                      const MoveArgs<W>&, hipStream_t);      \
+  template void launch_block<W, 8192>(const int32_t*, int,                  \
+                                      const MoveArgs<W>&, hipStream_t);      \
   template void launch_hub<W>(const int32_t*, int, const int64_t*, int64_t, \
                               const int64_t*, int32_t*, W*, double*,         \
                               int32_t*, double*, int64_t*, int32_t*,         \
