@@ -212,3 +212,32 @@ def test_distributed_louvain_early_term(et):
     outs = run_dist(2, _w_louvain, "karate",
                     {"early_term": et, "et_delta": 0.5})
     assert outs[0][0] > 0.15  # converges to a sane modularity (ET trades quality)
+
+
+def _w_louvain_seeded(rank, world, seed):
+    g = rmat_graph(7, 12, seed=seed)
+    dg = _shard(g, rank, world)
+    res = louvain(dg, Comm(), LouvainConfig(backend="torch"))
+    return res.modularity, res.communities.cpu()
+
+
+@pytest.mark.parametrize("seed", [11, 12, 13])
+def test_fuzz_p_invariance(seed):
+    """Random R-MAT graphs: 3-rank result equals single-rank result exactly
+    (protocol fuzz over different cut structures)."""
+    single = louvain(single_partition(rmat_graph(7, 12, seed=seed)), Comm(),
+                     LouvainConfig(backend="torch"))
+    outs = run_dist(3, _w_louvain_seeded, seed)
+    comms = torch.cat([o[1] for o in outs])
+    assert abs(outs[0][0] - single.modularity) < 1e-9
+    assert torch.equal(comms, single.communities)
+
+
+def test_world5_uneven_partition():
+    """World size that divides nothing evenly (uneven ranges, tiny shards)."""
+    single = louvain(single_partition(karate_graph()), Comm(),
+                     LouvainConfig(backend="torch"))
+    outs = run_dist(5, _w_louvain, "karate", {})
+    comms = torch.cat([o[1] for o in outs])
+    assert abs(outs[0][0] - single.modularity) < 1e-9
+    assert torch.equal(comms, single.communities)
