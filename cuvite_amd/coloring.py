@@ -115,10 +115,12 @@ def distance1_coloring(dg: DistGraph, comm: Comm,
     return colors, num_colors
 
 
-def check_coloring(dg: DistGraph, comm: Comm, colors: torch.Tensor) -> int:
-    """Count same-color adjacent pairs, excluding the overflow class
-    (ref distCheckColoring, coloring.cpp:447-593). Returns global conflicts
-    among properly-colored vertices."""
+def check_coloring(dg: DistGraph, comm: Comm, colors: torch.Tensor,
+                   exclude_color: int = -1) -> int:
+    """Count same-color adjacent pairs (ref distCheckColoring,
+    coloring.cpp:447-593). `exclude_color`: skip pairs in this class (the
+    uncolored-leftover overflow class is legitimately conflicting). Returns
+    the global conflict count."""
     dev = dg.g.device
     nv = dg.nv
     halo = build_halo(dg, comm)
@@ -128,5 +130,8 @@ def check_coloring(dg: DistGraph, comm: Comm, colors: torch.Tensor) -> int:
     call = torch.cat([colors, ghost_colors])
     gid_all = torch.cat([torch.arange(dg.base, dg.bound, device=dev), halo.ghosts])
     not_self = gid_all[tails] != (seg + dg.base)
-    conflicts = int((not_self & (call[tails] == colors[seg])).sum())
+    same = not_self & (call[tails] == colors[seg])
+    if exclude_color >= 0:
+        same &= colors[seg] != exclude_color
+    conflicts = int(same.sum())
     return int(comm.allreduce_scalar(float(conflicts)))
