@@ -242,3 +242,15 @@ def test_world5_uneven_partition():
     comms = torch.cat([o[1] for o in outs])
     assert abs(outs[0][0] - single.modularity) < 1e-9
     assert torch.equal(comms, single.communities)
+
+
+@pytest.mark.parametrize("kwargs", [
+    {"coloring": True, "max_colors": 6, "threshold_scaling": True},
+    {"ordering": True, "max_colors": 6, "early_term": 1},
+])
+def test_variant_combinations(kwargs):
+    """Flag combinations users mix (-c -i / -d -t 1) run to completion across
+    ranks and produce sane modularity."""
+    outs = run_dist(2, _w_louvain, "rmat", kwargs)
+    assert outs[0][0] > 0.1
+    assert outs[0][2] >= 1  # at least one iteration
