@@ -23,12 +23,23 @@ class Comm:
     """Thin communicator. world_size 1 (no process group) short-circuits
     every operation, so single-process runs need no init."""
 
-    def __init__(self, device: Optional[torch.device] = None):
+    def __init__(self, device: Optional[torch.device] = None,
+                 transport: Optional[str] = None):
         self.active = dist.is_available() and dist.is_initialized()
         self.rank = dist.get_rank() if self.active else 0
         self.world = dist.get_world_size() if self.active else 1
         self.device = device if device is not None else torch.device("cpu")
         self.backend = dist.get_backend() if self.active else None
+        # var-size exchange transport (ref compile-time matrix: Isend/Irecv
+        # vs MPI_Alltoallv, louvain.cpp:2634-2682): "p2p" = grouped
+        # batch_isend_irecv (default; all-pairs xGMI), "alltoall" = one
+        # fused ncclAllToAllv-style call (nccl backend only; gloo has no
+        # all_to_all). Runtime-selectable: CUVITE_TRANSPORT=alltoall.
+        if transport is None:
+            transport = os.environ.get("CUVITE_TRANSPORT", "p2p")
+        if transport == "alltoall" and self.backend != "nccl":
+            transport = "p2p"
+        self.transport = transport
 
     # -------------------------------------------------------- collectives --
 
@@ -104,6 +115,14 @@ class Comm:
                                   device=self.device)
             mat = self.allgather_counts(counts)
             recv_counts = [int(mat[p][me]) for p in range(self.world)]
+        if self.transport == "alltoall":
+            inp = torch.cat([send[p].contiguous() for p in range(self.world)])
+            out = torch.empty(sum(recv_counts), dtype=dtype,
+                              device=self.device)
+            dist.all_to_all_single(
+                out, inp, output_split_sizes=recv_counts,
+                input_split_sizes=[int(s.numel()) for s in send])
+            return list(torch.split(out, recv_counts))
         recv = [torch.empty(recv_counts[p], dtype=dtype, device=self.device)
                 for p in range(self.world)]
         ops = []
