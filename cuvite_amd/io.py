@@ -114,22 +114,30 @@ def write_dist_graph(path: str, shards):
 
 
 def load_ground_truth(path: str, zero_based: bool = True) -> torch.Tensor:
-    """Load an LFR-style ground-truth file: one `<vertex> <community>` pair per
-    line (ref loadGroundTruthFile, louvain.cpp:3272-3304). Returns community[v]."""
+    """Load a ground-truth/community file. Two formats are accepted:
+    `<vertex> <community>` pairs per line (the LFR ground-truth layout, ref
+    loadGroundTruthFile louvain.cpp:3272-3304) or one community id per line
+    indexed by line number (the reference's own dump format,
+    main.cpp:521-550). Returns community[v]."""
     data = np.loadtxt(path, dtype=np.int64, ndmin=2)
-    v, c = data[:, 0], data[:, 1]
+    if data.shape[1] >= 2:
+        v, c = data[:, 0], data[:, 1]
+        if not zero_based:
+            v = v - 1
+            c = c - 1
+        out = np.zeros(int(v.max()) + 1 if v.size else 0, dtype=np.int64)
+        out[v] = c
+        return torch.from_numpy(out)
+    c = data[:, 0]
     if not zero_based:
-        v = v - 1
         c = c - 1
-    out = np.zeros(int(v.max()) + 1 if v.size else 0, dtype=np.int64)
-    out[v] = c
-    return torch.from_numpy(out)
+    return torch.from_numpy(c.copy())
 
 
 def write_communities(path: str, comm: torch.Tensor):
-    """Dump `<vertex> <community>` per line (ref main.cpp:521-550 writes the
-    composed global clustering to `<input>.communities`)."""
+    """Dump one community id per line, line number = vertex id (exactly the
+    reference's <input>.communities format, main.cpp:536-538)."""
     c = comm.cpu().numpy()
     with open(path, "w") as f:
         for v in range(len(c)):
-            f.write(f"{v} {int(c[v])}\n")
+            f.write(f"{int(c[v])}\n")
