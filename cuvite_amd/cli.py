@@ -233,8 +233,10 @@ def main(argv=None) -> int:
     if comm.rank == 0:
         for lvl, q in enumerate(res.modularity_per_level):
             print(f"Level {lvl}: modularity = {q:.6f}")
-        # TEPS per the reference definition (main.cpp:448,509):
-        # edges traversed = global directed edge count x total iterations
+        # TEPS following the reference definition (main.cpp:448,509) with
+        # one fix: the reference multiplies each level's ne by the CUMULATIVE
+        # iteration count (inflating later levels); we use the original ne x
+        # total iterations, which is conservative (coarse levels are smaller)
         teps = ne_global * res.total_iters / t_total if t_total > 0 else 0.0
         print(f"Final modularity: {res.modularity:.6f}")
         print(f"Phases: {res.phases}  Iterations: {res.total_iters}")
