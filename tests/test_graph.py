@@ -78,3 +78,36 @@ def test_rmat_graph_symmetric():
     dg = single_partition(g)
     vdeg = dg.local_degree_sum()
     assert float(vdeg.sum()) == pytest.approx(2 * float(g.weights.sum()) / 2, rel=1e-12)
+
+
+def test_rgg_p_invariant():
+    """RGG shards for P=2 union to exactly the P=1 edge list (coords are pure
+    functions of the global vertex id; ref makes the graph P-dependent)."""
+    import numpy as np
+    from cuvite_amd.generators import rgg_local_edges
+    s1, d1, w1 = rgg_local_edges(256, 0, 1, seed=3, random_edge_percent=2.0)
+    parts = [rgg_local_edges(256, r, 2, seed=3, random_edge_percent=2.0)
+             for r in range(2)]
+    s2 = np.concatenate([p[0] for p in parts])
+    d2 = np.concatenate([p[1] for p in parts])
+    a = sorted(zip(s1.tolist(), d1.tolist()))
+    b = sorted(zip(s2.tolist(), d2.tolist()))
+    assert a == b
+
+
+def test_coarsen_weight_conservation_rmat():
+    import torch
+    from cuvite_amd.coarsen import coarsen
+    from cuvite_amd.generators import rmat_graph
+    from cuvite_amd.graph import single_partition
+    from cuvite_amd.parallel import Comm
+    g = rmat_graph(9, 8, seed=6)
+    dg = single_partition(g)
+    cvect = torch.arange(g.nv, dtype=torch.int64) // 7
+    new_dg, renum = coarsen(dg, Comm(torch.device("cpu")), cvect)
+    assert float(new_dg.g.weights.sum()) == pytest.approx(
+        float(g.weights.sum()), rel=1e-12)
+    # renumbered ids are a contiguous range
+    ids = renum(torch.unique(cvect))
+    assert torch.equal(torch.sort(ids).values,
+                       torch.arange(ids.numel(), dtype=torch.int64))
