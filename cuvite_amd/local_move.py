@@ -150,8 +150,12 @@ def local_move_torch(inp: MoveInputs):
 
     cand = ~own
     cv, cy, eiy = gv[cand], gc[cand], gsum[cand]
-    gain = (2.0 * (eiy - eix[cv])
-            - 2.0 * inp.v_degree[cv] * (inp.comm_degree[cy] - ax[cv]) * inp.constant)
+    # gain arithmetic always in fp64 (the HIP kernels do the same regardless
+    # of the weight dtype, so fp32-weight tie-breaks stay aligned)
+    gain = (2.0 * (eiy.to(torch.float64) - eix[cv].to(torch.float64))
+            - 2.0 * inp.v_degree[cv].to(torch.float64)
+            * (inp.comm_degree[cy].to(torch.float64)
+               - ax[cv].to(torch.float64)) * inp.constant)
 
     target = cc.clone()
     if cv.numel():
