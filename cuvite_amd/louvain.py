@@ -105,10 +105,11 @@ class PhaseState:
         arange = torch.arange(dg.base, dg.bound, device=dev)
         self.curr_comm = arange.clone()
         self.past_comm = arange.clone()
-        # constant = 1/(2m) (ref distCalcConstantForSecondTerm)
+        # constant = 1/(2m) (ref distCalcConstantForSecondTerm; guard the
+        # edgeless-graph case the reference would divide by zero on)
         tw = float(self.v_degree.to(torch.float64).sum())
         tw = comm.allreduce_scalar(tw)
-        self.constant = 1.0 / tw
+        self.constant = 1.0 / tw if tw > 0.0 else 0.0
         # ET state
         self.active = torch.ones(nv, dtype=torch.bool, device=dev)
         self.stable_count = torch.zeros(nv, dtype=torch.int16, device=dev)

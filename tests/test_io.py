@@ -70,3 +70,29 @@ def test_ground_truth_io(tmp_path):
             f.write(f"{v+1} {int(comm[v])+1}\n")
     gt1 = load_ground_truth(p, zero_based=False)
     assert torch.equal(gt1, comm)
+
+
+def test_empty_and_tiny_graphs(tmp_path):
+    """Degenerate shapes: single vertex, no edges."""
+    import numpy as np
+    from cuvite_amd.graph import Graph, single_partition
+    from cuvite_amd.louvain import louvain, LouvainConfig
+    from cuvite_amd.parallel import Comm
+
+    g = Graph(torch.tensor([0, 0], dtype=torch.int64),
+              torch.zeros(0, dtype=torch.int64),
+              torch.zeros(0, dtype=torch.float64))
+    # isolated single vertex: modularity 0, stays singleton
+    res = louvain(single_partition(g), Comm(torch.device("cpu")),
+                  LouvainConfig(backend="torch", one_phase=True))
+    assert res.communities.tolist() == [0]
+
+    p = tmp_path / "tiny.bin"
+    write_graph(str(p), g)
+    g2 = load_graph(str(p))
+    assert g2.nv == 1 and g2.ne == 0
+
+
+def test_load_missing_file_raises(tmp_path):
+    with pytest.raises(FileNotFoundError):
+        load_graph(str(tmp_path / "nope.bin"))
