@@ -254,3 +254,24 @@ def test_variant_combinations(kwargs):
     outs = run_dist(2, _w_louvain, "rmat", kwargs)
     assert outs[0][0] > 0.1
     assert outs[0][2] >= 1  # at least one iteration
+
+
+def _w_lfr_coloring(rank, world):
+    from cuvite_amd.generators import lfr_dist_graph
+    dg, truth = lfr_dist_graph(1500, rank, world, mu=0.3, seed=5)
+    res = louvain(dg, Comm(), LouvainConfig(backend="torch", coloring=True,
+                                            max_colors=6))
+    return res.modularity, res.communities.cpu(), truth
+
+
+def test_lfr_coloring_4rank_acceptance():
+    """BASELINE config 4 analog on CPU: LFR + coloring-ordered moves across
+    4 ranks recovers the planted communities."""
+    from cuvite_amd.compare import compare_communities
+    outs = run_dist(4, _w_lfr_coloring)
+    pred = torch.cat([o[1] for o in outs])
+    truth = outs[0][2]
+    m = compare_communities(truth, pred)
+    assert m["recall"] > 0.8
+    assert m["f_score"] > 0.7
+    assert outs[0][0] > 0.3  # healthy modularity
