@@ -96,3 +96,35 @@ def test_empty_and_tiny_graphs(tmp_path):
 def test_load_missing_file_raises(tmp_path):
     with pytest.raises(FileNotFoundError):
         load_graph(str(tmp_path / "nope.bin"))
+
+
+def test_convert_edge_list(tmp_path):
+    """Text edge list -> Vite binary -> Louvain end to end."""
+    from cuvite_amd.convert import convert_edge_list
+    from cuvite_amd.generators import karate_graph
+
+    # write karate as a 1-based weighted edge list (each undirected edge once)
+    g = karate_graph()
+    lines = ["# karate club"]
+    seen = set()
+    for u in range(g.nv):
+        for k in range(int(g.rowptr[u]), int(g.rowptr[u + 1])):
+            v = int(g.tails[k])
+            if (v, u) in seen or (u, v) in seen:
+                continue
+            seen.add((u, v))
+            lines.append(f"{u + 1} {v + 1} {float(g.weights[k])}")
+    src = tmp_path / "karate.txt"
+    src.write_text("\n".join(lines) + "\n")
+    out = tmp_path / "karate.bin"
+    g2 = convert_edge_list(str(src), str(out), one_based=True)
+    assert g2.nv == g.nv and g2.ne == g.ne
+    g3 = load_graph(str(out))
+    assert torch.equal(g3.rowptr, g.rowptr)
+
+    from cuvite_amd.graph import single_partition
+    from cuvite_amd.louvain import louvain, LouvainConfig
+    from cuvite_amd.parallel import Comm
+    res = louvain(single_partition(g3), Comm(torch.device("cpu")),
+                  LouvainConfig(backend="torch"))
+    assert res.modularity == pytest.approx(0.408695, abs=1e-6)
