@@ -179,3 +179,38 @@ def test_hub_moves_sorted_chunked_matches_unchunked():
         ops._HUB_SORT_CHUNK = old
     assert torch.equal(t1, t2)
     assert torch.allclose(cw1, cw2)
+
+
+def test_fuzz_oracle_vs_pydict():
+    """Randomized graphs + random community states: the vectorized oracle and
+    the dict-based golden transcription must agree exactly."""
+    for seed in range(12):
+        g = torch.Generator().manual_seed(900 + seed)
+        nv = int(torch.randint(2, 24, (1,), generator=g))
+        ne = int(torch.randint(1, 80, (1,), generator=g))
+        src = torch.randint(0, nv, (ne,), generator=g)
+        dst = torch.randint(0, nv, (ne,), generator=g)
+        # unit weights: exact arithmetic in any order
+        w = torch.ones(ne, dtype=torch.float64)
+        from cuvite_amd.graph import Graph
+        gr = Graph.from_edge_tuples(nv, torch.cat([src, dst]),
+                                    torch.cat([dst, src]),
+                                    torch.cat([w, w]))
+        curr = torch.randint(0, nv, (nv,), generator=g)
+        size = torch.zeros(nv, dtype=torch.int64)
+        size.index_add_(0, curr, torch.ones(nv, dtype=torch.int64))
+        deg = gr.rowptr[1:] - gr.rowptr[:-1]
+        vdeg = torch.zeros(nv, dtype=torch.float64)
+        seg = torch.repeat_interleave(torch.arange(nv), deg)
+        vdeg.index_add_(0, seg, gr.weights)
+        cdeg = torch.zeros(nv, dtype=torch.float64)
+        cdeg.index_add_(0, curr, vdeg)
+        tot = float(vdeg.sum())
+        inp = MoveInputs(gr.rowptr, gr.tails.to(torch.int32), gr.weights,
+                         curr.to(torch.int32), vdeg, size, cdeg,
+                         torch.arange(nv, dtype=torch.int64),
+                         1.0 / tot if tot else 0.0)
+        t1, c1 = local_move_torch(inp)
+        t2, c2 = local_move_pydict(inp)
+        assert torch.equal(t1.to(torch.int64), t2.to(torch.int64)), seed
+        assert torch.allclose(c1, c2), seed
