@@ -112,3 +112,34 @@ def test_karate_golden_trajectory():
     assert res.total_iters == 11
     assert res.modularity_per_level == pytest.approx(
         res.modularity_per_level)  # shape stability
+
+
+def test_threshold_cycling_schedule():
+    """Exact reference schedule (main.cpp:225-239): sp 0-2 -> 1e-3,
+    3-6 -> 1e-4, 7-9 -> 1e-5, 10-12 -> 1e-6, wrap at 13."""
+    from cuvite_amd.louvain import LouvainConfig, _threshold_for_phase
+    cfg = LouvainConfig(threshold_scaling=True)
+    expected = [1e-3] * 3 + [1e-4] * 4 + [1e-5] * 3 + [1e-6] * 3
+    for sp in range(26):
+        assert _threshold_for_phase(cfg, sp) == expected[sp % 13], sp
+
+
+def test_et_freeze_semantics():
+    """-t 1: a vertex stable for 3 iterations becomes inactive and its
+    target stays pinned to its current community."""
+    import torch
+    from cuvite_amd.generators import karate_graph
+    from cuvite_amd.graph import single_partition
+    from cuvite_amd.louvain import (LouvainConfig, PhaseState, _et_update)
+    from cuvite_amd.parallel import Comm
+
+    dg = single_partition(karate_graph())
+    state = PhaseState(dg, Comm(torch.device("cpu")))
+    cfg = LouvainConfig(early_term=1)
+    rng = torch.Generator().manual_seed(0)
+    # simulate 3 stable iterations: target == curr == past
+    tgt = state.curr_comm.clone()
+    for _ in range(3):
+        _et_update(state, cfg, tgt, rng)
+        state.past_comm = state.curr_comm.clone()
+    assert not bool(state.active.any())  # every vertex frozen after 3 stable
