@@ -111,3 +111,29 @@ def test_coarsen_weight_conservation_rmat():
     ids = renum(torch.unique(cvect))
     assert torch.equal(torch.sort(ids).values,
                        torch.arange(ids.numel(), dtype=torch.int64))
+
+
+def test_partition_owner_lookup():
+    part = Partition.contiguous(100, 7)
+    v = torch.arange(100)
+    owners = part.owner(v)
+    for p in range(7):
+        b, e = part.base(p), part.bound(p)
+        assert (owners[b:e] == p).all()
+    assert part.owner_one(0) == 0
+    assert part.owner_one(99) == 6
+
+
+def test_timers_accumulate():
+    import time as _t
+    from cuvite_amd.utils.timers import Timer, Timers
+    with Timer() as t:
+        _t.sleep(0.01)
+    assert t.elapsed >= 0.01
+    ts = Timers()
+    for _ in range(3):
+        with ts("x"):
+            _t.sleep(0.002)
+    assert ts.count["x"] == 3
+    assert ts.acc["x"] >= 0.006
+    assert "x=" in ts.summary()
