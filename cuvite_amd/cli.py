@@ -197,7 +197,10 @@ def main(argv=None) -> int:
         sys.stdout = open(f"dat.out.{comm.rank}", "w")
 
     t0 = time.perf_counter()
-    dg = _ingest(args, comm)
+    try:
+        dg = _ingest(args, comm)
+    except FileNotFoundError as e:
+        sys.exit(f"Error opening graph file: {e.filename}")
     t_ingest = time.perf_counter() - t0
     ne_global = comm.allreduce_scalar(float(dg.ne))
     if comm.rank == 0:
@@ -266,9 +269,12 @@ def main(argv=None) -> int:
                 write_communities(out_path, allc)
                 print(f"Wrote communities to {out_path}")
             if args.ground_truth or builtin_truth is not None:
-                truth = builtin_truth if builtin_truth is not None else \
-                    load_ground_truth(args.ground_truth,
-                                      zero_based=not args.one_based)
+                try:
+                    truth = builtin_truth if builtin_truth is not None else \
+                        load_ground_truth(args.ground_truth,
+                                          zero_based=not args.one_based)
+                except FileNotFoundError as e:
+                    sys.exit(f"Error opening ground truth file: {e.filename}")
                 m = compare_communities(truth, allc)
                 print(f"Ground truth: precision={m['precision']:.4f} "
                       f"recall={m['recall']:.4f} f-score={m['f_score']:.4f} "
