@@ -275,3 +275,45 @@ def test_lfr_coloring_4rank_acceptance():
     assert m["recall"] > 0.8
     assert m["f_score"] > 0.7
     assert outs[0][0] > 0.3  # healthy modularity
+
+
+def _w_count_rounds(rank, world):
+    """Count torch.distributed calls issued by ONE louvain iteration."""
+    import torch.distributed as dist
+    from cuvite_amd.louvain import PhaseState, LouvainConfig, _one_sweep, \
+        _modularity, _pick_move_fn
+    g = rmat_graph(8, 8, seed=3)
+    dg = _shard(g, rank, world)
+    comm = Comm()
+    counts = {"all_reduce": 0, "all_gather": 0, "batch": 0}
+    orig = (dist.all_reduce, dist.all_gather, dist.batch_isend_irecv)
+
+    def wrap(name, fn):
+        def inner(*a, **k):
+            counts[name] += 1
+            return fn(*a, **k)
+        return inner
+    dist.all_reduce = wrap("all_reduce", orig[0])
+    dist.all_gather = wrap("all_gather", orig[1])
+    dist.batch_isend_irecv = wrap("batch", orig[2])
+    try:
+        state = PhaseState(dg, comm)
+        cfg = LouvainConfig(backend="torch")
+        move_fn = _pick_move_fn(cfg, torch.device("cpu"))
+        counts = {k: 0 for k in counts}  # reset after setup
+        _one_sweep(state, cfg, move_fn, None)
+        _modularity(state)
+    finally:
+        dist.all_reduce, dist.all_gather, dist.batch_isend_irecv = orig
+    return counts
+
+
+def test_per_iteration_comm_rounds():
+    """Protocol-cost guard: one iteration must stay at <= 2 allgathers,
+    <= 5 grouped p2p rounds and exactly 1 allreduce (regressions here are
+    silent perf bugs at 8 GPUs)."""
+    outs = run_dist(2, _w_count_rounds)
+    c = outs[0]
+    assert c["all_reduce"] == 1, c          # modularity only
+    assert c["all_gather"] <= 2, c          # count negotiations
+    assert c["batch"] <= 5, c               # label xchg + info + deltas
