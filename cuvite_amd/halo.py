@@ -175,18 +175,18 @@ def push_remote_deltas(ctx: HaloContext, gids: torch.Tensor,
     gids, d_size, d_degree = gids[order], d_size[order], d_degree[order]
     parts = dg.partition.parts.to(dev)
     offs = torch.searchsorted(gids, parts)
-    sp = [gids[offs[p]:offs[p + 1]] for p in range(comm.world)]
-    got_ids = comm.all_to_all_v(sp)
-    counts = [2 * int(g.numel()) for g in got_ids]
-    # fused payload per peer: [delta-size, delta-degree] as fp64
-    fused = [torch.cat([d_size[offs[p]:offs[p + 1]].to(torch.float64),
+    # ONE p2p round: [ids, delta-size, delta-degree] per peer as fp64
+    # (vertex ids and integer deltas are exact in fp64 below 2^53)
+    fused = [torch.cat([gids[offs[p]:offs[p + 1]].to(torch.float64),
+                        d_size[offs[p]:offs[p + 1]].to(torch.float64),
                         d_degree[offs[p]:offs[p + 1]].to(torch.float64)])
              for p in range(comm.world)]
-    got = comm.all_to_all_v(fused, recv_counts=counts)
+    got = comm.all_to_all_v(fused)
     for p in range(comm.world):
-        n = got_ids[p].numel()
+        n = got[p].numel() // 3
         if n == 0:
             continue
-        li = got_ids[p] - dg.base
-        local_size.index_add_(0, li, got[p][:n].to(torch.int64))
-        scatter_add_(local_degree, li, got[p][n:].to(local_degree.dtype))
+        li = got[p][:n].to(torch.int64) - dg.base
+        local_size.index_add_(0, li, got[p][n:2 * n].to(torch.int64))
+        scatter_add_(local_degree, li,
+                     got[p][2 * n:].to(local_degree.dtype))
