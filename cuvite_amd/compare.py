@@ -32,10 +32,15 @@ def _best_f_mean(truth: np.ndarray, pred: np.ndarray):
     f = 2 * prec * rec / np.maximum(prec + rec, 1e-300)
     best = np.zeros(nt)
     np.maximum.at(best, ti, f)
-    # precision/recall of the pair achieving the best F per truth community
-    order = np.argsort(f, kind="stable")
-    best_pair = np.zeros(nt, dtype=np.int64)
-    best_pair[ti[order]] = order  # last write per ti = argmax f
+    # precision/recall of the pair achieving the best F per truth community:
+    # sort pairs by (ti, f) and take each ti-segment's last entry (the argmax;
+    # explicit reduction — duplicate-fancy-index assignment order is
+    # unspecified in NumPy). Every truth community appears in >=1 pair, so
+    # the segment ends enumerate all nt communities in order.
+    order = np.lexsort((f, ti))
+    ti_s = ti[order]
+    seg_end = np.flatnonzero(np.r_[ti_s[1:] != ti_s[:-1], True])
+    best_pair = order[seg_end]
     return float(best.mean()), float(prec[best_pair].mean()), \
         float(rec[best_pair].mean())
 

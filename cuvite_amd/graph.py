@@ -127,7 +127,13 @@ class Graph:
         (histogram + atomic-cursor placement; torch.argsort is capped at
         INT_MAX elements and a lexicographic sort is not needed — all CSR
         consumers are row-order-invariant). The CPU path sorts by (src, dst)
-        for determinism."""
+        for determinism.
+
+        Determinism note: the GPU builder leaves
+        row-internal edge order nondeterministic, so fp reductions over rows
+        (row_sum, the hub-merge cumsum) are reproducible only per CSR
+        instance, not bit-for-bit across builds of the same edge list; the
+        CPU path's sorted CSR is the bitwise-reproducible oracle."""
         if src.is_cuda:
             from . import ops
             rowptr, tails, weights = ops.csr_from_edges(nv, base, src, dst, w)

@@ -93,26 +93,51 @@ def _side_stream(dev):
 
 
 _HUB_SORT_CHUNK = 1 << 30  # max edges per sort batch (torch op INT_MAX cap)
+_hub_groups_cache: dict = {}
+
+
+def _hub_groups(inp, hubs, hdeg):
+    """Phase-static chunking of the hub list into groups whose edge totals
+    stay under the sort cap. The group tensors are cached per phase (keyed by
+    the rowptr storage) so _hub_static's identity-keyed cache hits every
+    iteration — fresh hubs[m] subsets each call would rebuild the multi-GB
+    flattened hub adjacency per iteration."""
+    key = (inp.rowptr.data_ptr(), hubs.data_ptr(), hubs.numel())
+    hit = _hub_groups_cache.get(key)
+    if hit is not None and hit[0] is inp.rowptr and hit[1] is hubs:
+        return hit[2]
+    total = int(hdeg.sum())
+    if total <= _HUB_SORT_CHUNK or hubs.numel() <= 1:
+        groups = [(hubs, hdeg, None)]
+    else:
+        cum = torch.cumsum(hdeg, dim=0)
+        gidx = torch.div(cum - 1, _HUB_SORT_CHUNK, rounding_mode="floor")
+        groups = []
+        for g in range(int(gidx[-1]) + 1):
+            m = gidx == g
+            if bool(m.any()):
+                groups.append((hubs[m].contiguous(), hdeg[m].contiguous(), m))
+    if len(_hub_groups_cache) > 4:
+        _hub_groups_cache.clear()
+    _hub_groups_cache[key] = (inp.rowptr, hubs, groups)
+    return groups
 
 
 def _hub_moves_sorted(inp, hubs, hdeg):
-    """Chunk wrapper: split the hub list into groups whose edge totals stay
-    under the torch sort cap; each hub's candidates are independent.
-    Returns (target_dense int32 [nhub], wcc [nhub]) aligned with `hubs`."""
-    total = int(hdeg.sum())
-    if total <= _HUB_SORT_CHUNK or hubs.numel() <= 1:
-        return _hub_moves_sorted_one(inp, hubs, hdeg)
-    cum = torch.cumsum(hdeg, dim=0)
-    group = torch.div(cum - 1, _HUB_SORT_CHUNK, rounding_mode="floor")
+    """Chunk wrapper: split the hub list into phase-static groups whose edge
+    totals stay under the torch sort cap; each hub's candidates are
+    independent. Returns (target_dense int32 [nhub], wcc [nhub]) aligned
+    with `hubs`."""
+    groups = _hub_groups(inp, hubs, hdeg)
+    if len(groups) == 1:
+        return _hub_moves_sorted_one(inp, groups[0][0], groups[0][1])
     tgt = torch.empty(hubs.numel(), dtype=torch.int32, device=hubs.device)
     wcc = torch.empty(hubs.numel(), dtype=inp.weights.dtype,
                       device=hubs.device)
-    for gidx in range(int(group[-1]) + 1):
-        m = group == gidx
-        if bool(m.any()):
-            t, w = _hub_moves_sorted_one(inp, hubs[m], hdeg[m])
-            tgt[m] = t
-            wcc[m] = w
+    for hubs_g, hdeg_g, m in groups:
+        t, w = _hub_moves_sorted_one(inp, hubs_g, hdeg_g)
+        tgt[m] = t
+        wcc[m] = w
     return tgt, wcc
 
 
@@ -179,7 +204,12 @@ def _hub_static(inp, hubs, hdeg):
         extra = None
     data = (seg, tails_h, wts, selfloop, extra)
     if len(_hub_static_cache) > 4:
-        _hub_static_cache.clear()
+        # evict only OTHER phases' entries: all chunk groups of the current
+        # phase must stay resident together (one entry per group)
+        stale = [k for k, v in _hub_static_cache.items()
+                 if v[0] is not inp.rowptr]
+        for k in stale:
+            del _hub_static_cache[k]
     _hub_static_cache[key] = (inp.rowptr, hubs, data)
     return data
 

@@ -231,6 +231,11 @@ std::vector<at::Tensor> hub_candidates(at::Tensor tails_flat,
   TORCH_CHECK(tails_flat.scalar_type() == at::kInt);
   TORCH_CHECK(seg_flat.scalar_type() == at::kInt);
   const int64_t n = tails_flat.numel();
+  // reduce_by_key writes its unique count as unsigned int; the argmax kernel
+  // reads it through an int32 pointer — enforce the range here instead of
+  // relying on the Python-side chunk cap
+  TORCH_CHECK(n < (int64_t)INT32_MAX,
+              "hub candidate batch exceeds int32 count range; chunk the hubs");
   const int nseg = (int)(eoffs.numel() - 1);
   auto stream = at::hip::getCurrentHIPStream().stream();
   int end_bit = 1;
@@ -295,6 +300,8 @@ std::vector<at::Tensor> hub_moves(
   TORCH_CHECK(hubs_i32.scalar_type() == at::kInt);
   TORCH_CHECK(hub_self.scalar_type() == at::kDouble);
   const int64_t n = tails_flat.numel();
+  TORCH_CHECK(n < (int64_t)INT32_MAX,
+              "hub candidate batch exceeds int32 count range; chunk the hubs");
   const int nhub = (int)hubs_i32.numel();
   const int64_t C = comm_degree.numel();
   auto stream = at::hip::getCurrentHIPStream().stream();

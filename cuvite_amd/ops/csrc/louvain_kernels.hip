@@ -288,7 +288,13 @@ __global__ __launch_bounds__(BLOCK) void lv_move_block(
 // 2*deg rounded to a power of two per hub (288 GB HBM: sized from the
 // graph, unlike the reference's hard-coded 2x2.7 GB buffers,
 // GpuGraph.cu:61-64). Every probe loop is bounded by cap as a hang guard
-// (sets `overflow` instead of spinning).
+// (sets `overflow` instead of spinning). Overflow is impossible by
+// construction: caps are 2*(deg+1) rounded up to a power of two
+// (ops/__init__.py _buckets_for), so distinct keys <= deg < cap/2 and a
+// linear probe always finds an empty slot; the flag exists solely so a
+// memory-corruption bug degrades to a wrong answer instead of a GPU hang,
+// which is why the host discards it on the (env-gated, experimental)
+// CUVITE_HUB_HIP path.
 // ---------------------------------------------------------------------------
 
 constexpr int HUB_SPLITS = 16;
