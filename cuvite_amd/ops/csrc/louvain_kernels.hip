@@ -17,13 +17,17 @@
 //                                (48 KB -> 3 blocks/CU; the 8192-slot table
 //                                is 96 KB and caps residency at 1 block/CU)
 //   block B: deg in (2048, 4096] one 256-thread block/vertex, 8192-slot LDS
-//   hub    : deg > 4096          DEFAULT: radix-sort + segmented-reduction
-//                                path (torch/rocPRIM, ops/__init__.py;
-//                                rocPRIM variant = hub_moves binding).
-//                                The hash-table pipeline below is kept
-//                                behind CUVITE_HUB_HIP (global tables past
-//                                the 4 MB XCD L2 measured pathologically
-//                                slow; profiles/hub_pathology_and_s26.md)
+//   hub    : deg > 4096          DEFAULT: hub_moves binding (rocPRIM
+//                                narrow-bit segmented radix sort +
+//                                reduce_by_key + wave-per-hub argmax) —
+//                                2.9x faster than the torch global-sort
+//                                fallback at s26 (A/B in profiles/;
+//                                CUVITE_HUB_SEGSORT=0 restores the
+//                                fallback). The hash-table pipeline below
+//                                is kept behind CUVITE_HUB_HIP (global
+//                                tables past the 4 MB XCD L2 measured
+//                                pathologically slow;
+//                                profiles/hub_pathology_and_s26.md)
 //
 // All gain arithmetic is fp64 regardless of the weight dtype so trajectories
 // match the fp64 CPU oracle (tie-break on equal gains -> smaller GLOBAL id).
