@@ -60,6 +60,13 @@ class LouvainResult:
     total_iters: int
     times: dict = field(default_factory=dict)
     modularity_per_level: list = field(default_factory=list)
+    levels: list = field(default_factory=list)  # per-phase {ne_global, iters,
+    #   modularity, cluster_s, rebuild_s} — feeds the reference TEPS
+    #   definition (teps += ne * iters per phase, main.cpp:448)
+
+    @property
+    def teps_numerator(self) -> float:
+        return sum(lv["ne_global"] * lv["iters"] for lv in self.levels)
 
 
 def _threshold_for_phase(cfg: LouvainConfig, short_phase: int) -> float:
@@ -407,6 +414,7 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
     short_phase = 0
     tot_iters = 0
     mods = []
+    levels: list = []
     level = dg
     times = {"coloring": 0.0, "clustering": 0.0, "rebuild": 0.0}
 
@@ -422,12 +430,19 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
                                                     n_hash=max(1, cfg.max_colors // 2))
             times["coloring"] += time.perf_counter() - t0
 
+        ne_global = comm.allreduce_scalar(float(level.ne)) \
+            if comm.world > 1 else float(level.ne)
         t0 = time.perf_counter()
         curr_mod, cvect, iters, phase_halo = run_phase(
             level, comm, cfg, curr_mod, threshold,
             colors=colors, num_colors=num_colors)
-        times["clustering"] += time.perf_counter() - t0
+        t_cluster = time.perf_counter() - t0
+        times["clustering"] += t_cluster
         tot_iters += iters
+        level_entry = {"ne_global": ne_global, "iters": iters,
+                       "modularity": curr_mod, "cluster_s": t_cluster,
+                       "rebuild_s": 0.0}
+        levels.append(level_entry)
 
         if (curr_mod - prev_mod) > threshold:
             # compose: originals currently assigned to level vertices; level
@@ -440,12 +455,14 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
             level, renum = coarsen(level, comm, cvect, halo=phase_halo)
             # cvect-composed orig_assign holds OLD comm gids; renumber them
             orig_assign = renum(orig_assign)
-            times["rebuild"] += time.perf_counter() - t0
+            level_entry["rebuild_s"] = time.perf_counter() - t0
+            times["rebuild"] += level_entry["rebuild_s"]
         else:
             if cfg.threshold_scaling and not cfg.one_phase and phase < 10:
                 curr_mod2, cvect, iters, _ = run_phase(level, comm, cfg,
                                                        curr_mod, 1.0e-6)
                 tot_iters += iters
+                level_entry["iters"] += iters
                 if (curr_mod2 - curr_mod) > 1.0e-6:
                     orig_assign = remap_labels(level, comm, orig_assign, cvect)
                     mods.append(curr_mod2)
@@ -463,4 +480,4 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
     times["total"] = time.perf_counter() - t_start
     final_mod = mods[-1] if mods else max(prev_mod, curr_mod)
     return LouvainResult(final_mod, orig_assign, phase + 1, tot_iters,
-                         times, mods)
+                         times, mods, levels)
