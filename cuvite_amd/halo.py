@@ -160,6 +160,40 @@ def fetch_remote_comm_info(ctx: HaloContext, remote_gids: torch.Tensor,
     return sizes, degrees
 
 
+def fetch_comm_info_lists(ctx: HaloContext, reqs: List[torch.Tensor],
+                          local_size: torch.Tensor,
+                          local_degree: torch.Tensor):
+    """Fetch (size, degree) of remote communities from their owners, with the
+    requests already grouped per owner rank (reqs[p] = gids owned by p, any
+    order). Returns per-peer lists (sizes[p] int64, degrees[p] W) aligned
+    with reqs[p]. Collective: every rank must call with world-length lists.
+    Ref: rounds 2-3 of fillRemoteCommunities (louvain.cpp:2688-2959); unlike
+    fetch_remote_comm_info this variant imposes no global-sort requirement
+    on the request set (the phase-persistent universe in louvain.densify is
+    append-ordered)."""
+    comm, dg = ctx.comm, ctx.dg
+    W = local_degree.dtype
+    got = comm.all_to_all_v(reqs)
+    # one fused reply per peer: [size-bits, degree-bits] as fp64 payload
+    # (sizes are exact in fp64 up to 2^53; halves the p2p rounds over xGMI)
+    reply = []
+    for p in range(comm.world):
+        if p == comm.rank or got[p].numel() == 0:
+            reply.append(torch.empty(0, dtype=torch.float64,
+                                     device=local_degree.device))
+            continue
+        li = got[p] - dg.base
+        reply.append(torch.cat([local_size[li].to(torch.float64),
+                                local_degree[li].to(torch.float64)]))
+    req_counts = [2 * int(r.numel()) for r in reqs]
+    back = comm.all_to_all_v(reply, recv_counts=req_counts)
+    sizes = [back[p][:back[p].numel() // 2].to(torch.int64)
+             for p in range(comm.world)]
+    degrees = [back[p][back[p].numel() // 2:].to(W)
+               for p in range(comm.world)]
+    return sizes, degrees
+
+
 def push_remote_deltas(ctx: HaloContext, gids: torch.Tensor,
                        d_size: torch.Tensor, d_degree: torch.Tensor,
                        local_size: torch.Tensor, local_degree: torch.Tensor):

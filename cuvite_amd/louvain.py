@@ -23,8 +23,8 @@ from typing import List, Optional
 import torch
 
 from .graph import DistGraph
-from .halo import (build_halo, exchange_ghost_labels,
-                   fetch_remote_comm_info, push_remote_deltas)
+from .halo import (build_halo, exchange_ghost_labels, fetch_comm_info_lists,
+                   push_remote_deltas)
 from .local_move import MoveInputs, local_move_torch, modularity_parts
 from .ops import scatter_add_
 from .parallel import Comm
@@ -121,11 +121,66 @@ class PhaseState:
         self.active = torch.ones(nv, dtype=torch.bool, device=dev)
         self.stable_count = torch.zeros(nv, dtype=torch.int16, device=dev)
         self.move_prob = torch.ones(nv, dtype=torch.float32, device=dev)
+        # phase-persistent dense community space (see densify)
+        self._lgids: Optional[torch.Tensor] = None
+        self._parts_dev: Optional[torch.Tensor] = None
+        self._runiv: Optional[torch.Tensor] = None
+        self._prev_labels: Optional[torch.Tensor] = None
+
+    def _local_gids(self) -> torch.Tensor:
+        if self._lgids is None:
+            self._lgids = torch.arange(self.dg.base, self.dg.bound,
+                                       device=self.dg.g.device)
+        return self._lgids
+
+    def _univ_lookup(self, rem: torch.Tensor):
+        """Position in the (append-ordered) universe for each gid in `rem`,
+        plus a found mask (positions are garbage where not found)."""
+        K = self._runiv_sorted.numel()
+        if K == 0:
+            z = torch.zeros(rem.numel(), dtype=torch.bool, device=rem.device)
+            return torch.zeros(rem.numel(), dtype=torch.int64,
+                               device=rem.device), z
+        idx = torch.searchsorted(self._runiv_sorted, rem).clamp(max=K - 1)
+        found = self._runiv_sorted[idx] == rem
+        return self._runiv_perm[idx], found
+
+    def _univ_append(self, new: torch.Tensor):
+        """Append new remote community gids to the universe (dense ids are
+        append-ordered so existing ids stay stable); rebuild the sorted
+        lookup index (only runs on growth — rare after iteration 1)."""
+        dev = new.device
+        if self._parts_dev is None:
+            self._parts_dev = self.dg.partition.parts.to(dev)
+        owner = torch.searchsorted(self._parts_dev, new, right=True) - 1
+        self._runiv = torch.cat([self._runiv, new])
+        self._univ_owner = torch.cat([self._univ_owner, owner])
+        self._univ_size = torch.cat([
+            self._univ_size,
+            torch.zeros(new.numel(), dtype=torch.int64, device=dev)])
+        self._univ_degree = torch.cat([
+            self._univ_degree,
+            torch.zeros(new.numel(), dtype=self.local_degree.dtype,
+                        device=dev)])
+        self._runiv_sorted, self._runiv_perm = torch.sort(self._runiv)
 
     def densify(self, ghost_comm: torch.Tensor):
-        """Map global labels to the per-iteration dense community space.
-        Returns (curr_dense [nv+ng] int32, remote_gids sorted, comm_size,
-        comm_degree, comm_gid) with remote info fetched from owners."""
+        """Map global labels to the dense community space.
+        Returns (curr_dense [nv+ng] int32, remote_gids (append-ordered
+        universe), comm_size, comm_degree, comm_gid) with remote info
+        fetched from owners for every REFERENCED remote community.
+
+        Phase-persistent design (SURVEY section 7 "biggest systems win"; the
+        reference rebuilds its dense remap on the host every iteration,
+        louvain_cuda.cu:2260-2378): remote community gids are interned into
+        an append-only universe, so per-iteration work is (a) an O(n)
+        changed-label compare against the previous iteration's labels,
+        (b) a searchsorted lookup for the changed ones only, and (c) an O(n)
+        scatter for the referenced-set mask — NO per-iteration torch.unique /
+        sort over nv+ng labels. Universe entries that are no longer
+        referenced keep stale aggregates; the move kernels only ever read
+        communities adjacent to some vertex, which are referenced and fresh
+        this iteration."""
         dg, dev = self.dg, self.dg.g.device
         base, bound, nv = dg.base, dg.bound, dg.nv
         if self.comm.world == 1:
@@ -133,21 +188,70 @@ class PhaseState:
             dense = (self.curr_comm - base).to(torch.int32)
             empty = torch.empty(0, dtype=torch.int64, device=dev)
             return (dense, empty, self.local_size, self.local_degree,
-                    torch.arange(base, bound, device=dev))
+                    self._local_gids())
         all_labels = torch.cat([self.curr_comm, ghost_comm])
-        is_local = (all_labels >= base) & (all_labels < bound)
-        remote_gids = torch.unique(all_labels[~is_local])
-        r_size, r_degree = fetch_remote_comm_info(
-            self.halo, remote_gids, self.local_size, self.local_degree)
-        dense = torch.empty_like(all_labels)
-        dense[is_local] = all_labels[is_local] - base
-        if remote_gids.numel():
-            dense[~is_local] = nv + torch.searchsorted(remote_gids,
-                                                       all_labels[~is_local])
-        comm_size = torch.cat([self.local_size, r_size])
-        comm_degree = torch.cat([self.local_degree, r_degree])
-        comm_gid = torch.cat([torch.arange(base, bound, device=dev), remote_gids])
-        return dense.to(torch.int32), remote_gids, comm_size, comm_degree, comm_gid
+        n = all_labels.numel()
+        if self._runiv is None:
+            i64 = dict(dtype=torch.int64, device=dev)
+            self._runiv = torch.empty(0, **i64)
+            self._runiv_sorted = self._runiv
+            self._runiv_perm = torch.empty(0, **i64)
+            self._univ_owner = torch.empty(0, **i64)
+            self._univ_size = torch.empty(0, **i64)
+            self._univ_degree = torch.empty(
+                0, dtype=self.local_degree.dtype, device=dev)
+            self._dense = torch.empty(n, dtype=torch.int32, device=dev)
+
+        if self._prev_labels is not None and self._prev_labels.numel() == n:
+            changed = (all_labels != self._prev_labels).nonzero(
+                as_tuple=True)[0]
+            lab_c = all_labels[changed]
+        else:
+            changed = None
+            lab_c = all_labels
+
+        if lab_c.numel():
+            is_local = (lab_c >= base) & (lab_c < bound)
+            rem = lab_c[~is_local]
+            pos, found = self._univ_lookup(rem)
+            if not bool(found.all()):
+                self._univ_append(torch.unique(rem[~found]))
+                pos, _ = self._univ_lookup(rem)
+            dense_c = torch.empty(lab_c.numel(), dtype=torch.int32,
+                                  device=dev)
+            dense_c[is_local] = (lab_c[is_local] - base).to(torch.int32)
+            if rem.numel():
+                dense_c[~is_local] = (nv + pos).to(torch.int32)
+            if changed is None:
+                self._dense = dense_c
+            else:
+                self._dense[changed] = dense_c
+        self._prev_labels = all_labels
+
+        # referenced remote communities this iteration (mask in universe
+        # order), then one request/reply round with their owners
+        K = self._runiv.numel()
+        ref_mask = torch.zeros(K, dtype=torch.bool, device=dev)
+        d64 = self._dense.to(torch.int64)
+        rr = d64[d64 >= nv] - nv
+        ref_mask[rr] = True
+        world = self.comm.world
+        reqs, positions = [], []
+        for p in range(world):
+            sel = (ref_mask & (self._univ_owner == p)).nonzero(
+                as_tuple=True)[0]
+            reqs.append(self._runiv[sel])
+            positions.append(sel)
+        sizes_l, degrees_l = fetch_comm_info_lists(
+            self.halo, reqs, self.local_size, self.local_degree)
+        for p in range(world):
+            if positions[p].numel():
+                self._univ_size[positions[p]] = sizes_l[p]
+                self._univ_degree[positions[p]] = degrees_l[p]
+        comm_size = torch.cat([self.local_size, self._univ_size])
+        comm_degree = torch.cat([self.local_degree, self._univ_degree])
+        comm_gid = torch.cat([self._local_gids(), self._runiv])
+        return self._dense, self._runiv, comm_size, comm_degree, comm_gid
 
     def apply_moves(self, target_gid: torch.Tensor, remote_gids: torch.Tensor):
         """Apply community size/degree deltas for vertices that moved
