@@ -46,14 +46,18 @@ def _hash_scalar(a: int, seed: int) -> int:
     return int(_hash(torch.tensor([a], dtype=torch.int64), seed)[0])
 
 
-def distance1_coloring(dg: DistGraph, comm: Comm,
-                       n_hash: int = 4) -> Tuple[torch.Tensor, int]:
+def distance1_coloring(dg: DistGraph, comm: Comm, n_hash: int = 4,
+                       halo=None) -> Tuple[torch.Tensor, int]:
     """Returns (colors int64 [nv] in [0, num_colors), num_colors).
-    Uncolored leftovers are assigned the last class num_colors-1."""
+    Uncolored leftovers are assigned the last class num_colors-1.
+    `halo`: optional prebuilt HaloContext — the phase loop passes the one
+    PhaseState will reuse, so -c/-d runs do ghost discovery once per phase
+    instead of twice (round-1 weak item 5)."""
     dev = dg.g.device
     nv = dg.nv
     base = dg.base
-    halo = build_halo(dg, comm)
+    if halo is None:
+        halo = build_halo(dg, comm)
     tails = halo.tails_dense.to(torch.int64)
     seg = torch.repeat_interleave(torch.arange(nv, device=dev), dg.g.degrees())
     gid_all = torch.cat([torch.arange(base, dg.bound, device=dev), halo.ghosts])
@@ -116,14 +120,15 @@ def distance1_coloring(dg: DistGraph, comm: Comm,
 
 
 def check_coloring(dg: DistGraph, comm: Comm, colors: torch.Tensor,
-                   exclude_color: int = -1) -> int:
+                   exclude_color: int = -1, halo=None) -> int:
     """Count same-color adjacent pairs (ref distCheckColoring,
     coloring.cpp:447-593). `exclude_color`: skip pairs in this class (the
     uncolored-leftover overflow class is legitimately conflicting). Returns
     the global conflict count."""
     dev = dg.g.device
     nv = dg.nv
-    halo = build_halo(dg, comm)
+    if halo is None:
+        halo = build_halo(dg, comm)
     tails = halo.tails_dense.to(torch.int64)
     seg = torch.repeat_interleave(torch.arange(nv, device=dev), dg.g.degrees())
     ghost_colors = exchange_ghost_labels(halo, colors)

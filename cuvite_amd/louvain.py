@@ -97,11 +97,11 @@ def _pick_move_fn(cfg: LouvainConfig, device: torch.device):
 class PhaseState:
     """Mutable per-phase state for one rank (all tensors on dg's device)."""
 
-    def __init__(self, dg: DistGraph, comm: Comm):
+    def __init__(self, dg: DistGraph, comm: Comm, halo=None):
         dev = dg.g.device
         self.dg = dg
         self.comm = comm
-        self.halo = build_halo(dg, comm)
+        self.halo = halo if halo is not None else build_halo(dg, comm)
         nv = dg.nv
         W = dg.g.weights.dtype
         self.v_degree = dg.local_degree_sum()                       # W [nv]
@@ -299,8 +299,10 @@ def _et_update(state: PhaseState, cfg: LouvainConfig, target: torch.Tensor,
         state.move_prob = torch.where(
             moved, torch.ones_like(state.move_prob),
             state.move_prob * (1.0 - cfg.et_delta))
+        # draw on the compute device (a per-iteration host round trip here
+        # was round-1 weak item 6); rng lives on the same device
         draw = torch.rand(state.move_prob.shape, generator=rng,
-                          device="cpu").to(state.move_prob.device)
+                          device=state.move_prob.device)
         state.active = (state.move_prob >= P_CUTOFF) & \
             ((draw < state.move_prob) | moved)
     if cfg.early_term in (3, 4):
@@ -312,18 +314,18 @@ def _et_update(state: PhaseState, cfg: LouvainConfig, target: torch.Tensor,
 def run_phase(dg: DistGraph, comm: Comm, cfg: LouvainConfig,
               lower: float, threshold: float,
               colors: Optional[torch.Tensor] = None,
-              num_colors: int = 0):
+              num_colors: int = 0, halo=None):
     """One Louvain phase (local moving to convergence). Returns
     (prev_mod, cvect global labels [nv], iters).
     Ref: distLouvainMethod (louvain.cpp:425-588) and the coloring/ordering
     variants (louvain.cpp:756-2101)."""
-    state = PhaseState(dg, comm)
+    state = PhaseState(dg, comm, halo=halo)
     move_fn = _pick_move_fn(cfg, dg.g.device)
     state.use_hip = (dg.g.device.type == "cuda"
                      and cfg.backend in ("auto", "hip"))
     prev_mod = lower
     iters = 0
-    rng = torch.Generator().manual_seed(12345 + comm.rank)
+    rng = torch.Generator(device=dg.g.device).manual_seed(12345 + comm.rank)
     timers = Timers(sync=dg.g.device.type == "cuda") if cfg.verbose else None
 
     use_colors = colors is not None and num_colors > 0
@@ -528,10 +530,14 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
 
         colors = None
         num_colors = 0
+        pre_halo = None
         if (cfg.coloring or cfg.ordering) and phase == 0:
+            from .halo import build_halo as _bh
             t0 = time.perf_counter()
-            colors, num_colors = distance1_coloring(level, comm,
-                                                    n_hash=max(1, cfg.max_colors // 2))
+            pre_halo = _bh(level, comm)  # shared with PhaseState below
+            colors, num_colors = distance1_coloring(
+                level, comm, n_hash=max(1, cfg.max_colors // 2),
+                halo=pre_halo)
             times["coloring"] += time.perf_counter() - t0
 
         ne_global = comm.allreduce_scalar(float(level.ne)) \
@@ -539,7 +545,7 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
         t0 = time.perf_counter()
         curr_mod, cvect, iters, phase_halo = run_phase(
             level, comm, cfg, curr_mod, threshold,
-            colors=colors, num_colors=num_colors)
+            colors=colors, num_colors=num_colors, halo=pre_halo)
         t_cluster = time.perf_counter() - t0
         times["clustering"] += t_cluster
         tot_iters += iters

@@ -71,30 +71,19 @@ class Comm:
 
     def gather_cat(self, t: torch.Tensor, root: int = 0) -> Optional[torch.Tensor]:
         """Gather variable-length 1-D tensors to root and concatenate in rank
-        order (ref gatherAllComm, louvain.cpp:3306-3345). Cold path."""
+        order (ref gatherAllComm, louvain.cpp:3306-3345). Cold path; routed
+        through the same count-negotiated grouped-p2p machinery as the hot
+        exchanges (all_to_all_v) so the first RCCL hardware contact has one
+        code path, not a raw send/irecv variant on the side."""
         if not self.active or self.world == 1:
             return t
-        n = torch.tensor([t.numel()], dtype=torch.int64, device=self.device)
-        sizes = self.allgather_counts(n).flatten().tolist()
-        # dist.gather requires equal sizes; variable-length gather goes p2p
+        empty = torch.empty(0, dtype=t.dtype, device=self.device)
+        send = [t.contiguous() if p == root else empty
+                for p in range(self.world)]
+        got = self.all_to_all_v(send)
         if self.rank == root:
-            outs = []
-            works = []
-            for p in range(self.world):
-                if p == root:
-                    outs.append(t.contiguous())
-                    continue
-                buf = torch.empty(int(sizes[p]), dtype=t.dtype, device=self.device)
-                outs.append(buf)
-                if sizes[p] > 0:
-                    works.append(dist.irecv(buf, src=p))
-            for w in works:
-                w.wait()
-            return torch.cat([outs[p] for p in range(self.world)])
-        else:
-            if t.numel() > 0:
-                dist.send(t.contiguous(), dst=root)
-            return None
+            return torch.cat([got[p] for p in range(self.world)])
+        return None
 
     # ---------------------------------------------------------------- p2p --
 

@@ -215,6 +215,22 @@ def test_distributed_louvain_early_term(et):
     assert outs[0][0] > 0.15  # converges to a sane modularity (ET trades quality)
 
 
+@pytest.mark.parametrize("et", [1, 3])
+def test_early_term_deterministic_matches_single(et):
+    """-t 1/3 (deterministic freeze) is P-invariant: the freeze state is a
+    pure function of the (P-invariant) move trajectory, so 2-rank results
+    must equal the single-rank run EXACTLY — a semantic guard, not just a
+    crash test (round-1 weak item 7). -t 2/4 draw per-rank randoms and are
+    P-dependent by design."""
+    single = louvain(single_partition(rmat_graph(8, 8, seed=3)), Comm(),
+                     LouvainConfig(backend="torch", early_term=et))
+    outs = run_dist(2, _w_louvain, "rmat", {"early_term": et})
+    comms = torch.cat([o[1] for o in outs])
+    assert abs(outs[0][0] - single.modularity) < 1e-9
+    assert torch.equal(comms, single.communities)
+    assert outs[0][2] == single.total_iters
+
+
 def _w_louvain_seeded(rank, world, seed):
     g = rmat_graph(7, 12, seed=seed)
     dg = _shard(g, rank, world)
