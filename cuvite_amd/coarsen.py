@@ -108,8 +108,21 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor,
     # fine edge list, so partial aggregation also bounds peak memory
     ne = dg.g.ne
     CH = 1 << 28
+    # chunk-aggregates are merged in one final sort while their total stays
+    # under the torch sort cap; past it (s30-per-rank territory) they are
+    # folded progressively so no single sort exceeds INT_MAX elements
+    MERGE_CAP = 3 << 29
     rowptr = dg.g.rowptr
     parts_s, parts_t, parts_w = [], [], []
+    pending = 0
+
+    def _fold():
+        nonlocal parts_s, parts_t, parts_w, pending
+        s, t, w = _aggregate(torch.cat(parts_s), torch.cat(parts_t),
+                             torch.cat(parts_w), gnc)
+        parts_s, parts_t, parts_w = [s], [t], [w]
+        pending = s.numel()
+
     for c0 in range(0, max(ne, 1), CH):
         c1 = min(c0 + CH, ne)
         if c1 <= c0:
@@ -124,10 +137,13 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor,
         parts_s.append(cs)
         parts_t.append(ct)
         parts_w.append(cw)
+        pending += cs.numel()
+        if pending > MERGE_CAP:
+            _fold()
     if parts_s:
-        s_agg, t_agg, w_agg = _aggregate(torch.cat(parts_s),
-                                         torch.cat(parts_t),
-                                         torch.cat(parts_w), gnc)
+        if len(parts_s) > 1:
+            _fold()
+        s_agg, t_agg, w_agg = parts_s[0], parts_t[0], parts_w[0]
     else:
         s_agg = torch.zeros(0, dtype=torch.int64, device=dev)
         t_agg = torch.zeros(0, dtype=torch.int64, device=dev)
