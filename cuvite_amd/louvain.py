@@ -258,6 +258,25 @@ class PhaseState:
         (ref 4-case update, louvain.cpp:2308-2376 + updateRemoteCommunities)."""
         dg = self.dg
         base, bound = dg.base, dg.bound
+        if getattr(self, "use_hip", False) and target_gid.is_cuda:
+            from . import ops
+            # fused local update, no host sync / compaction kernels
+            ops.apply_deltas_(target_gid, self.curr_comm, self.v_degree,
+                              base, bound, self.local_size,
+                              self.local_degree)
+            if self.comm.world == 1:
+                return
+            moved = target_gid != self.curr_comm
+            src = self.curr_comm[moved]
+            dst = target_gid[moved]
+            vdeg = self.v_degree[moved]
+            gids = torch.cat([src, dst])
+            dsize = torch.cat([-torch.ones_like(src), torch.ones_like(dst)])
+            ddeg = torch.cat([-vdeg, vdeg])
+            rem = (gids < base) | (gids >= bound)
+            push_remote_deltas(self.halo, gids[rem], dsize[rem], ddeg[rem],
+                               self.local_size, self.local_degree)
+            return
         moved = target_gid != self.curr_comm
         if not bool(moved.any()):
             # still must participate in the collective delta push

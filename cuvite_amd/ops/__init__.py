@@ -422,6 +422,17 @@ def scatter_add_(out: torch.Tensor, idx: torch.Tensor,
     return out
 
 
+def apply_deltas_(target: torch.Tensor, curr: torch.Tensor,
+                  v_degree: torch.Tensor, base: int, bound: int,
+                  local_size: torch.Tensor, local_degree: torch.Tensor):
+    """Fused post-sweep community-aggregate update: for every vertex whose
+    label changed, apply the +-1 size / +-degree deltas to LOCALLY-owned
+    communities with native atomics (one kernel; replaces the torch
+    mask/compact/index_add chain measured at ~40 ms/sweep at s26)."""
+    _require().apply_deltas_(target, curr, v_degree, base, bound,
+                             local_size, local_degree)
+
+
 def csr_from_edges(nv: int, base: int, src: torch.Tensor, dst: torch.Tensor,
                    w: torch.Tensor):
     """Sort-free device CSR assembly: (rowptr, tails, weights). Handles edge

@@ -59,7 +59,12 @@ template <typename W>
 void launch_hub_argmax(const int64_t*, const W*, const int32_t*, int64_t,
                        const int32_t*, int, const double*, const int32_t*,
                        const W*, const int64_t*, const W*, const int64_t*,
-                       double, int32_t*, W*, hipStream_t);
+                       double, double*, int64_t*, int32_t*, int32_t*, W*,
+                       hipStream_t);
+int hub_argmax_splits();
+template <typename W>
+void launch_apply_deltas(const int64_t*, const int64_t*, const W*, int64_t,
+                         int64_t, int64_t, int64_t*, W*, hipStream_t);
 
 }  // namespace cuvite
 
@@ -352,16 +357,49 @@ std::vector<at::Tensor> hub_moves(
                                uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
                                (unsigned int*)cnt.data_ptr<int32_t>(),
                                stream);
+    const int splits = cuvite::hub_argmax_splits();
+    auto p_gain = at::empty({(int64_t)nhub * splits},
+                            eoffs.options().dtype(at::kDouble));
+    auto p_gid = at::empty({(int64_t)nhub * splits},
+                           eoffs.options().dtype(at::kLong));
+    auto p_dense = at::empty({(int64_t)nhub * splits},
+                             tails_flat.options());
     cuvite::launch_hub_argmax<W>(
         uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
         cnt.data_ptr<int32_t>(), C, hubs_i32.data_ptr<int32_t>(), nhub,
         hub_self.data_ptr<double>(), curr_comm.data_ptr<int32_t>(),
         v_degree.data_ptr<W>(), comm_size.data_ptr<int64_t>(),
         comm_degree.data_ptr<W>(), comm_gid.data_ptr<int64_t>(), constant,
+        p_gain.data_ptr<double>(), p_gid.data_ptr<int64_t>(),
+        p_dense.data_ptr<int32_t>(),
         target_hub.data_ptr<int32_t>(), cw_hub.data_ptr<W>(), stream);
   });
   C10_HIP_CHECK(hipGetLastError());
   return {target_hub, cw_hub};
+}
+
+void apply_deltas_(at::Tensor target, at::Tensor curr, at::Tensor v_degree,
+                   int64_t base, int64_t bound, at::Tensor local_size,
+                   at::Tensor local_degree) {
+  CHECK_DEV(target); CHECK_CONT(target);
+  CHECK_DEV(curr); CHECK_CONT(curr);
+  CHECK_DEV(v_degree); CHECK_CONT(v_degree);
+  CHECK_DEV(local_size); CHECK_CONT(local_size);
+  CHECK_DEV(local_degree); CHECK_CONT(local_degree);
+  TORCH_CHECK(target.scalar_type() == at::kLong &&
+              curr.scalar_type() == at::kLong, "labels must be int64");
+  TORCH_CHECK(local_size.scalar_type() == at::kLong);
+  TORCH_CHECK(target.numel() == curr.numel() &&
+              target.numel() == v_degree.numel());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES(local_degree.scalar_type(), "apply_deltas", [&] {
+    cuvite::launch_apply_deltas<scalar_t>(
+        target.data_ptr<int64_t>(), curr.data_ptr<int64_t>(),
+        v_degree.data_ptr<scalar_t>(), target.numel(), base, bound,
+        local_size.data_ptr<int64_t>(), local_degree.data_ptr<scalar_t>(),
+        stream);
+  });
+  C10_HIP_CHECK(hipGetLastError());
 }
 
 at::Tensor row_sum(at::Tensor rowptr, at::Tensor weights) {
@@ -391,6 +429,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("csr_from_edges", &csr_from_edges,
         "sort-free CSR assembly on device (HIP)");
   m.def("row_sum", &row_sum, "per-row CSR weight sum (HIP)");
+  m.def("apply_deltas_", &apply_deltas_,
+        "fused community size/degree delta update for moved vertices (HIP)");
   m.def("hub_candidates", &hub_candidates,
         "segmented-sort + reduce_by_key hub candidate generation (rocPRIM)");
   m.def("hub_moves", &hub_moves,

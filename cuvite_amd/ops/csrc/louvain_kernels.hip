@@ -140,6 +140,17 @@ DEV_INLINE void scan_slots(const int32_t* keys, const W* vals, int cap,
 // vlist is padded to a multiple of (BLOCK/LANES) with -1.
 // ---------------------------------------------------------------------------
 
+// Smallest power of two >= 2*(deg+1), clamped to [lo, CAP]: distinct keys
+// <= deg keeps the load factor under 1/2 while init/probe/scan touch only
+// `cap` slots. rocprof at s26 showed the class-3 kernel (CAP 4096, typical
+// deg ~600) spending most of its time on table init + argmax scan of slots
+// that can never be occupied (profiles/round2_kernel_stats).
+DEV_INLINE int cap_for_degree(int64_t deg, int CAP, int lo) {
+  int cap = CAP;
+  while (cap > lo && (cap >> 2) >= (int)(deg + 1)) cap >>= 1;
+  return cap;
+}
+
 template <typename W, int LANES, int CAP, int BLOCK>
 __global__ __launch_bounds__(BLOCK) void lv_move_sub(
     const int32_t* __restrict__ vlist, int nlist,
@@ -161,12 +172,6 @@ __global__ __launch_bounds__(BLOCK) void lv_move_sub(
 
   int32_t* gkeys = keys + (size_t)group * CAP;
   W* gvals = vals + (size_t)group * CAP;
-#pragma unroll 4
-  for (int s = lane; s < CAP; s += LANES) {
-    gkeys[s] = EMPTY_KEY;
-    gvals[s] = (W)0;
-  }
-  __syncthreads();
 
   int64_t e0 = 0, e1 = 0;
   int32_t cc = 0;
@@ -175,24 +180,32 @@ __global__ __launch_bounds__(BLOCK) void lv_move_sub(
     e1 = rowptr[v + 1];
     cc = curr_comm[v];
   }
+  const int cap = cap_for_degree(e1 - e0, CAP, LANES);
+#pragma unroll 4
+  for (int s = lane; s < cap; s += LANES) {
+    gkeys[s] = EMPTY_KEY;
+    gvals[s] = (W)0;
+  }
+  __syncthreads();
+
   double selfloop = 0.0;
   for (int64_t e = e0 + lane; e < e1; e += LANES) {
     int32_t t = tails[e];
     W w = weights[e];
     if (t == v) selfloop += (double)w;
-    table_insert(gkeys, gvals, CAP, curr_comm[t], w);
+    table_insert(gkeys, gvals, cap, curr_comm[t], w);
   }
   __syncthreads();
 
   selfloop = sum_reduce<LANES>(selfloop);
   Best best{0.0, 0, cc};
   if (v >= 0 && e0 != e1) {
-    W wcc = table_probe(gkeys, gvals, CAP, cc);  // counter[cc]
+    W wcc = table_probe(gkeys, gvals, cap, cc);  // counter[cc]
     double eix = (double)wcc - __shfl(selfloop, 0, LANES);
     double vdeg = (double)v_degree[v];
     double ax = (double)comm_degree[cc] - vdeg;
     best.gid = comm_gid[cc];
-    scan_slots(gkeys, gvals, CAP, lane, LANES, cc, eix, ax, vdeg, constant,
+    scan_slots(gkeys, gvals, cap, lane, LANES, cc, eix, ax, vdeg, constant,
                comm_degree, comm_gid, best);
     best_reduce<LANES>(best);
     if (lane == 0) {
@@ -232,20 +245,21 @@ __global__ __launch_bounds__(BLOCK) void lv_move_block(
   const int tid = threadIdx.x;
   const int wave = tid / 64, lane = tid % 64;
 
-  for (int s = tid; s < CAP; s += BLOCK) {
+  const int64_t e0 = rowptr[v], e1 = rowptr[v + 1];
+  const int cap = cap_for_degree(e1 - e0, CAP, BLOCK);
+  for (int s = tid; s < cap; s += BLOCK) {
     keys[s] = EMPTY_KEY;
     vals[s] = (W)0;
   }
   __syncthreads();
 
-  const int64_t e0 = rowptr[v], e1 = rowptr[v + 1];
   const int32_t cc = curr_comm[v];
   double selfloop = 0.0;
   for (int64_t e = e0 + tid; e < e1; e += BLOCK) {
     int32_t t = tails[e];
     W w = weights[e];
     if (t == v) selfloop += (double)w;
-    table_insert(keys, vals, CAP, curr_comm[t], w);
+    table_insert(keys, vals, cap, curr_comm[t], w);
   }
   selfloop = sum_reduce<64>(selfloop);
   if (lane == 0) red_self[wave] = selfloop;
@@ -255,12 +269,12 @@ __global__ __launch_bounds__(BLOCK) void lv_move_block(
 #pragma unroll
   for (int i = 0; i < WAVES; i++) self_total += red_self[i];
 
-  W wcc = table_probe(keys, vals, CAP, cc);
+  W wcc = table_probe(keys, vals, cap, cc);
   double eix = (double)wcc - self_total;
   double vdeg = (double)v_degree[v];
   double ax = (double)comm_degree[cc] - vdeg;
   Best best{0.0, comm_gid[cc], cc};
-  scan_slots(keys, vals, CAP, tid, BLOCK, cc, eix, ax, vdeg, constant,
+  scan_slots(keys, vals, cap, tid, BLOCK, cc, eix, ax, vdeg, constant,
              comm_degree, comm_gid, best);
   best_reduce<64>(best);
   if (lane == 0) {
@@ -505,6 +519,41 @@ __global__ void scatter_add_kernel(W* __restrict__ out,
 }
 
 // ---------------------------------------------------------------------------
+// Fused community-aggregate update after a move sweep (ref 4-case update,
+// louvain.cpp:2308-2376): one pass over the vertices applies the +-1 size
+// and +-v_degree deltas for every moved vertex directly, with native int64/
+// fp atomics. Replaces a 6-op torch chain (moved mask -> .any() host sync ->
+// boolean compactions -> cat -> index_add_ -> scatter_add) that rocprof
+// measured at ~40 ms/sweep at s26 (profiles/round2_kernel_stats). Only
+// LOCALLY-owned community labels are applied here; remotely-owned deltas are
+// compacted by the caller for the RCCL push (world>1 only).
+// ---------------------------------------------------------------------------
+
+template <typename W>
+__global__ void apply_deltas_kernel(
+    const int64_t* __restrict__ target, const int64_t* __restrict__ curr,
+    const W* __restrict__ v_degree, int64_t nv, int64_t base, int64_t bound,
+    int64_t* __restrict__ local_size, W* __restrict__ local_degree) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+       i += stride) {
+    const int64_t src = curr[i];
+    const int64_t dst = target[i];
+    if (src == dst) continue;
+    const W d = v_degree[i];
+    if (src >= base && src < bound) {
+      atomicAdd((unsigned long long*)&local_size[src - base],
+                (unsigned long long)(-1ll));
+      unsafeAtomicAdd(&local_degree[src - base], (W)(-d));
+    }
+    if (dst >= base && dst < bound) {
+      atomicAdd((unsigned long long*)&local_size[dst - base], 1ull);
+      unsafeAtomicAdd(&local_degree[dst - base], d);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Device-side CSR assembly (no sort; replaces torch.argsort which is capped
 // at INT_MAX elements): degree histogram + atomic-cursor placement.
 // Reference analog: processGraphData (utils.cpp:10-87) done host-side there.
@@ -585,49 +634,60 @@ __global__ void pack_key64_kernel(const int32_t* __restrict__ keys,
     key64[i] = (int64_t)seg_flat[i] * C + keys[i];
 }
 
-// One wave per hub over its (sorted) candidate range: wcc pass, then the
-// exact-tie-break dQ argmax (same Best semantics as the class kernels).
+// dQ argmax over the per-hub (sorted, deduped) candidate ranges produced by
+// reduce_by_key. Two facts make this cheap: (a) keys are UNIQUE, so the
+// weight to the own community wcc is a single binary search, not a range
+// scan; (b) the argmax over a hub's range is split across ARGMAX_SPLITS
+// blocks so an R-MAT mega-hub (10^6 candidates) cannot serialize on one
+// wave — rocprof at s26 showed the wave-per-hub version at 20-45 ms/sweep
+// from exactly that imbalance (profiles/round2_kernel_stats).
+
+constexpr int ARGMAX_SPLITS = 8;
+
+DEV_INLINE int64_t lower_bound64(const int64_t* __restrict__ a, int64_t n,
+                                 int64_t key) {
+  int64_t lo = 0, hi = n;
+  while (lo < hi) {
+    const int64_t m = (lo + hi) >> 1;
+    if (a[m] < key) lo = m + 1; else hi = m;
+  }
+  return lo;
+}
+
 template <typename W, int BLOCK>
-__global__ __launch_bounds__(BLOCK) void hub_argmax_kernel(
+__global__ __launch_bounds__(BLOCK) void hub_argmax_part_kernel(
     const int64_t* __restrict__ uniq, const W* __restrict__ sums,
     const int32_t* __restrict__ cnt_ptr, int64_t C,
     const int32_t* __restrict__ hubs, int nhub,
     const double* __restrict__ hub_self,
     const int32_t* __restrict__ curr_comm, const W* __restrict__ v_degree,
-    const int64_t* __restrict__ comm_size, const W* __restrict__ comm_degree,
-    const int64_t* __restrict__ comm_gid, double constant,
-    int32_t* __restrict__ target_hub, W* __restrict__ cw_hub) {
+    const W* __restrict__ comm_degree, const int64_t* __restrict__ comm_gid,
+    double constant, double* __restrict__ p_gain,
+    int64_t* __restrict__ p_gid, int32_t* __restrict__ p_dense) {
   constexpr int WAVES = BLOCK / 64;
-  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  const int hidx = blockIdx.x * WAVES + wave;
-  if (hidx >= nhub) return;
+  __shared__ double red_gain[WAVES];
+  __shared__ int64_t red_gid[WAVES];
+  __shared__ int32_t red_dense[WAVES];
+  const int hidx = blockIdx.x / ARGMAX_SPLITS;
+  const int part = blockIdx.x % ARGMAX_SPLITS;
   const int64_t n = (int64_t)cnt_ptr[0];
   const int32_t v = hubs[hidx];
   const int32_t cc = curr_comm[v];
   const int64_t keylo = (int64_t)hidx * C;
-  const int64_t keyhi = keylo + C;
-  int64_t lo = 0, hi = n;
-  while (lo < hi) {
-    const int64_t m = (lo + hi) >> 1;
-    if (uniq[m] < keylo) lo = m + 1; else hi = m;
-  }
-  const int64_t r0 = lo;
-  hi = n;
-  while (lo < hi) {
-    const int64_t m = (lo + hi) >> 1;
-    if (uniq[m] < keyhi) lo = m + 1; else hi = m;
-  }
-  const int64_t r1 = lo;
-  double wcc = 0.0;
-  for (int64_t i = r0 + lane; i < r1; i += 64)
-    if ((int32_t)(uniq[i] - keylo) == cc) wcc += (double)sums[i];
-  wcc = sum_reduce<64>(wcc);
-  wcc = __shfl(wcc, 0, 64);
+  const int64_t r0 = lower_bound64(uniq, n, keylo);
+  const int64_t r1 = lower_bound64(uniq, n, keylo + C);
+  // unique keys: the own-community weight is one lookup
+  const int64_t pcc = lower_bound64(uniq, n, keylo + cc);
+  const double wcc =
+      (pcc < n && uniq[pcc] == keylo + cc) ? (double)sums[pcc] : 0.0;
   const double eix = wcc - hub_self[hidx];
   const double vdeg = (double)v_degree[v];
   const double ax = (double)comm_degree[cc] - vdeg;
+  const int64_t len = r1 - r0;
+  const int64_t s0 = r0 + part * len / ARGMAX_SPLITS;
+  const int64_t s1 = r0 + (part + 1) * len / ARGMAX_SPLITS;
   Best best{0.0, comm_gid[cc], cc};
-  for (int64_t i = r0 + lane; i < r1; i += 64) {
+  for (int64_t i = s0 + threadIdx.x; i < s1; i += BLOCK) {
     const int32_t y = (int32_t)(uniq[i] - keylo);
     if (y == cc) continue;
     const double eiy = (double)sums[i];
@@ -636,12 +696,55 @@ __global__ __launch_bounds__(BLOCK) void hub_argmax_kernel(
     best_combine(best, g, comm_gid[y], y);
   }
   best_reduce<64>(best);
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  if (lane == 0) {
+    red_gain[wave] = best.gain;
+    red_gid[wave] = best.gid;
+    red_dense[wave] = best.dense;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+#pragma unroll
+    for (int i = 1; i < WAVES; i++)
+      best_combine(best, red_gain[i], red_gid[i], red_dense[i]);
+    p_gain[blockIdx.x] = best.gain;
+    p_gid[blockIdx.x] = best.gid;
+    p_dense[blockIdx.x] = best.dense;
+  }
+}
+
+template <typename W, int BLOCK>
+__global__ __launch_bounds__(BLOCK) void hub_argmax_final_kernel(
+    const int64_t* __restrict__ uniq, const W* __restrict__ sums,
+    const int32_t* __restrict__ cnt_ptr, int64_t C,
+    const int32_t* __restrict__ hubs, int nhub,
+    const int32_t* __restrict__ curr_comm,
+    const int64_t* __restrict__ comm_size, const int64_t* __restrict__ comm_gid,
+    const double* __restrict__ p_gain, const int64_t* __restrict__ p_gid,
+    const int32_t* __restrict__ p_dense, int32_t* __restrict__ target_hub,
+    W* __restrict__ cw_hub) {
+  constexpr int WAVES = BLOCK / 64;
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const int hidx = blockIdx.x * WAVES + wave;
+  if (hidx >= nhub) return;
+  const int64_t n = (int64_t)cnt_ptr[0];
+  const int32_t v = hubs[hidx];
+  const int32_t cc = curr_comm[v];
+  Best best{0.0, comm_gid[cc], cc};
+  if (lane < ARGMAX_SPLITS) {
+    const int p = hidx * ARGMAX_SPLITS + lane;
+    best = Best{p_gain[p], p_gid[p], p_dense[p]};
+  }
+  best_reduce<64>(best);
   if (lane == 0) {
     int32_t tgt = best.dense;
     if (comm_size[tgt] == 1 && comm_size[cc] == 1 && best.gid > comm_gid[cc])
       tgt = cc;
     target_hub[hidx] = tgt;
-    cw_hub[hidx] = (W)wcc;
+    const int64_t keycc = (int64_t)hidx * C + cc;
+    const int64_t pcc = lower_bound64(uniq, n, keycc);
+    cw_hub[hidx] =
+        (pcc < n && uniq[pcc] == keycc) ? sums[pcc] : (W)0;
   }
 }
 
@@ -777,6 +880,25 @@ template void launch_scatter_add<float>(float*, const int64_t*, const float*,
 template void launch_scatter_add<double>(double*, const int64_t*,
                                          const double*, int64_t, hipStream_t);
 
+template <typename W>
+void launch_apply_deltas(const int64_t* target, const int64_t* curr,
+                         const W* v_degree, int64_t nv, int64_t base,
+                         int64_t bound, int64_t* local_size, W* local_degree,
+                         hipStream_t stream) {
+  if (nv == 0) return;
+  hipLaunchKernelGGL((apply_deltas_kernel<W>), dim3(grid_for(nv, 256)),
+                     dim3(256), 0, stream, target, curr, v_degree, nv, base,
+                     bound, local_size, local_degree);
+}
+template void launch_apply_deltas<float>(const int64_t*, const int64_t*,
+                                         const float*, int64_t, int64_t,
+                                         int64_t, int64_t*, float*,
+                                         hipStream_t);
+template void launch_apply_deltas<double>(const int64_t*, const int64_t*,
+                                          const double*, int64_t, int64_t,
+                                          int64_t, int64_t*, double*,
+                                          hipStream_t);
+
 void launch_degree_count(const int64_t* src, int64_t n, int64_t base,
                          int32_t* cnt, hipStream_t stream) {
   if (n == 0) return;
@@ -825,30 +947,39 @@ void launch_hub_argmax(const int64_t* uniq, const W* sums,
                        const int32_t* curr_comm, const W* v_degree,
                        const int64_t* comm_size, const W* comm_degree,
                        const int64_t* comm_gid, double constant,
+                       double* p_gain, int64_t* p_gid, int32_t* p_dense,
                        int32_t* target_hub, W* cw_hub, hipStream_t stream) {
   if (nhub == 0) return;
   constexpr int BLOCK = 256;
   constexpr int WAVES = BLOCK / 64;
-  hipLaunchKernelGGL((hub_argmax_kernel<W, BLOCK>),
-                     dim3((nhub + WAVES - 1) / WAVES), dim3(BLOCK), 0, stream,
+  hipLaunchKernelGGL((hub_argmax_part_kernel<W, BLOCK>),
+                     dim3(nhub * ARGMAX_SPLITS), dim3(BLOCK), 0, stream,
                      uniq, sums, cnt, C, hubs, nhub, hub_self, curr_comm,
-                     v_degree, comm_size, comm_degree, comm_gid, constant,
-                     target_hub, cw_hub);
+                     v_degree, comm_degree, comm_gid, constant, p_gain,
+                     p_gid, p_dense);
+  hipLaunchKernelGGL((hub_argmax_final_kernel<W, BLOCK>),
+                     dim3((nhub + WAVES - 1) / WAVES), dim3(BLOCK), 0, stream,
+                     uniq, sums, cnt, C, hubs, nhub, curr_comm, comm_size,
+                     comm_gid, p_gain, p_gid, p_dense, target_hub, cw_hub);
 }
 template void launch_hub_argmax<float>(const int64_t*, const float*,
                                        const int32_t*, int64_t,
                                        const int32_t*, int, const double*,
                                        const int32_t*, const float*,
                                        const int64_t*, const float*,
-                                       const int64_t*, double, int32_t*,
+                                       const int64_t*, double, double*,
+                                       int64_t*, int32_t*, int32_t*,
                                        float*, hipStream_t);
 template void launch_hub_argmax<double>(const int64_t*, const double*,
                                         const int32_t*, int64_t,
                                         const int32_t*, int, const double*,
                                         const int32_t*, const double*,
                                         const int64_t*, const double*,
-                                        const int64_t*, double, int32_t*,
+                                        const int64_t*, double, double*,
+                                        int64_t*, int32_t*, int32_t*,
                                         double*, hipStream_t);
+
+int hub_argmax_splits() { return ARGMAX_SPLITS; }
 
 template <typename W>
 void launch_row_sum(const int64_t* rowptr, const W* weights, int64_t nv,

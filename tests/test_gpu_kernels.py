@@ -270,12 +270,10 @@ def test_gpu_lfr_recovery():
     assert m["f_score"] > 0.6
 
 
-@pytest.mark.skipif(not os.environ.get("CUVITE_TEST_SEGSORT"),
-                    reason="experimental rocPRIM segsort hub path "
-                           "(set CUVITE_TEST_SEGSORT=1)")
-def test_hub_segsort_matches_default():
-    """CUVITE_HUB_SEGSORT=1 (rocPRIM segmented sort + reduce_by_key) must
-    match the default torch-sort hub path exactly on unit weights."""
+def test_hub_segsort_matches_torch_sort_fallback():
+    """The default rocPRIM segsort hub path (hub_moves binding, incl. the
+    split-block argmax) must match the torch global-sort fallback
+    (CUVITE_HUB_SEGSORT=0) exactly on unit weights."""
     from cuvite_amd import ops
     torch.manual_seed(11)
     nv = 8192
@@ -287,16 +285,46 @@ def test_hub_segsort_matches_default():
     g = Graph.from_edge_tuples(nv, src, dst, w)
     dev = torch.device("cuda:0")
     inp = _inputs(g, dev, "random", seed=5)
-    ops._bucket_cache.clear()
-    ops._hub_static_cache.clear()
-    t_ref, cw_ref = ops.local_move(inp)
-    os.environ["CUVITE_HUB_SEGSORT"] = "1"
+    os.environ["CUVITE_HUB_SEGSORT"] = "0"
     try:
         ops._bucket_cache.clear()
         ops._hub_static_cache.clear()
-        t_new, cw_new = ops.local_move(inp)
+        t_ref, cw_ref = ops.local_move(inp)
     finally:
         del os.environ["CUVITE_HUB_SEGSORT"]
-        ops._hub_static_cache.clear()
+    ops._bucket_cache.clear()
+    ops._hub_static_cache.clear()
+    t_new, cw_new = ops.local_move(inp)
+    ops._hub_static_cache.clear()
     assert torch.equal(t_ref, t_new)
     assert torch.allclose(cw_ref, cw_new)
+
+
+def test_apply_deltas_matches_torch():
+    """Fused apply_deltas_ kernel vs the plain torch index_add/scatter_add
+    reference, including remote (out-of-range) labels that must be ignored."""
+    from cuvite_amd import ops
+    dev = torch.device("cuda:0")
+    torch.manual_seed(7)
+    nv, base, bound = 5000, 1000, 6000
+    curr = torch.randint(0, 8000, (nv,), dtype=torch.int64, device=dev)
+    target = curr.clone()
+    m = torch.rand(nv, device=dev) < 0.4
+    target[m] = torch.randint(0, 8000, (int(m.sum()),), dtype=torch.int64,
+                              device=dev)
+    vdeg = torch.rand(nv, dtype=torch.float64, device=dev) * 3
+    size = torch.randint(1, 5, (bound - base,), dtype=torch.int64, device=dev)
+    degree = torch.rand(bound - base, dtype=torch.float64, device=dev)
+    size_ref, degree_ref = size.clone(), degree.clone()
+    # torch reference
+    moved = target != curr
+    gids = torch.cat([curr[moved], target[moved]])
+    ds = torch.cat([-torch.ones_like(curr[moved]),
+                    torch.ones_like(target[moved])])
+    dd = torch.cat([-vdeg[moved], vdeg[moved]])
+    loc = (gids >= base) & (gids < bound)
+    size_ref.index_add_(0, gids[loc] - base, ds[loc])
+    degree_ref.index_add_(0, gids[loc] - base, dd[loc])
+    ops.apply_deltas_(target, curr, vdeg, base, bound, size, degree)
+    assert torch.equal(size, size_ref)
+    assert torch.allclose(degree, degree_ref)
