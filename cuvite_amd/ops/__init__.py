@@ -39,12 +39,13 @@ _bucket_cache: dict = {}
 
 
 def _buckets_for(rowptr: torch.Tensor):
-    """Degree-class vertex lists + hub-table offsets for a CSR (static per
-    phase; cached by the rowptr storage)."""
+    """Degree-class vertex lists for a CSR (static per phase; cached by the
+    rowptr storage). Returns (vlists[5] for the LDS class kernels,
+    hubs64 int64 hub vertex list, hdeg int64 hub degrees)."""
     key = (rowptr.data_ptr(), rowptr.numel())
     hit = _bucket_cache.get(key)
     if hit is not None and hit[0] is rowptr:  # identity check: ptr reuse safe
-        return hit[1], hit[2], hit[3], hit[4], hit[5]
+        return hit[1], hit[2], hit[3]
     deg = rowptr[1:] - rowptr[:-1]
     b0, b1, b2, b3, b4 = _CLASS_BOUNDS
     vlists = [
@@ -53,34 +54,15 @@ def _buckets_for(rowptr: torch.Tensor):
         ((deg > b1) & (deg <= b2)).nonzero(as_tuple=True)[0].to(torch.int32),
         ((deg > b2) & (deg <= b3)).nonzero(as_tuple=True)[0].to(torch.int32),
         ((deg > b3) & (deg <= b4)).nonzero(as_tuple=True)[0].to(torch.int32),
-        (deg > b4).nonzero(as_tuple=True)[0].to(torch.int32),
     ]
-    hubs = vlists[5]
-    if hubs.numel():
-        hdeg = deg[hubs.to(torch.int64)]
-        caps = torch.pow(
-            2.0, torch.ceil(torch.log2((2 * (hdeg + 1)).to(torch.float64)))
-        ).to(torch.int64)
-        offsets = torch.zeros(hubs.numel() + 1, dtype=torch.int64,
-                              device=rowptr.device)
-        offsets[1:] = torch.cumsum(caps, dim=0)
-        eoffs = torch.zeros(hubs.numel() + 1, dtype=torch.int64,
-                            device=rowptr.device)
-        eoffs[1:] = torch.cumsum(hdeg, dim=0)
-        total_hub_edges = int(eoffs[-1])
-    else:
-        offsets = torch.zeros(1, dtype=torch.int64, device=rowptr.device)
-        eoffs = torch.zeros(1, dtype=torch.int64, device=rowptr.device)
-        total_hub_edges = 0
-    hubs64 = hubs.to(torch.int64)
+    hubs64 = (deg > b4).nonzero(as_tuple=True)[0]
+    hdeg = deg[hubs64]
     if len(_bucket_cache) > 8:
         _bucket_cache.clear()
-    _bucket_cache[key] = (rowptr, vlists, offsets, eoffs, total_hub_edges,
-                          hubs64)
-    return vlists, offsets, eoffs, total_hub_edges, hubs64
+    _bucket_cache[key] = (rowptr, vlists, hubs64, hdeg)
+    return vlists, hubs64, hdeg
 
 
-_pool_cache: dict = {}
 _side_streams: dict = {}
 
 
@@ -225,12 +207,14 @@ def _hub_static(inp, hubs, hdeg):
 
 
 def _hub_moves_sorted_one(inp, hubs, hdeg):
-    """Hub vertices (deg > 4096) via radix sort + segmented reduction instead
-    of the global hash-table pipeline: rocPRIM sort (torch.sort) of
-    (hub, community) keys, cumsum segment sums, and a vectorized exact-
-    tie-break argmax. Deterministic, atomic-free; the hash-table pipeline
-    (CUVITE_HUB_HIP=1) showed pathological slowdowns on tables past the 4 MB
-    XCD-L2 footprint (see profiles/ hang bisection)."""
+    """Hub vertices (deg > 4096) via radix sort + segmented reduction.
+    Default path: the fully-device rocPRIM pipeline (hub_moves binding).
+    Fallback (CUVITE_HUB_SEGSORT=0): torch.sort of packed (hub, community)
+    keys + cumsum segment sums + vectorized exact-tie-break argmax.
+    Deterministic, atomic-free. A round-1 global-memory hash-table pipeline
+    was deleted after showing pathological CAS slowdowns on tables past the
+    4 MB XCD-L2 footprint (profiles/hub_pathology_and_s26.md) and losing
+    the A/B to this pipeline (profiles/round2_hub_segsort_ab.md)."""
     dev = inp.rowptr.device
     nhub = hubs.numel()
     seg, tails_h, wts, selfloop, extra = _hub_static(inp, hubs, hdeg)
@@ -298,50 +282,29 @@ def _hub_moves_sorted_one(inp, hubs, hdeg):
 
 def local_move(inp):
     """HIP local-move iteration (see local_move.MoveInputs for semantics).
-    Returns (target dense comm ids [nv], cluster_weight [nv])."""
+    Returns (target dense comm ids [nv], cluster_weight [nv]).
+
+    Degree classes 0-4 run in the LDS hash-table kernels; hub vertices
+    (deg > 4096) go through the rocPRIM segsort pipeline (_hub_moves_sorted),
+    overlapped on separate HIP streams (disjoint vertex sets)."""
     ext = _require()
-    vlists, offsets, eoffs, n_hub_edges, hubs64 = _buckets_for(inp.rowptr)
+    vlists, hubs64, hdeg = _buckets_for(inp.rowptr)
     dev = inp.rowptr.device
-    hub_hip = bool(os.environ.get("CUVITE_HUB_HIP"))
-    if not hub_hip:
-        # hub class handled by the sort-based path after the HIP classes
-        n_pool = 0
-        pool_keys = torch.empty(0, dtype=torch.int32, device=dev)
-        pool_vals = torch.empty(0, dtype=inp.weights.dtype, device=dev)
-        vlists = vlists[:5] + [vlists[5][:0]]
-    else:
-        n_pool = int(offsets[-1])
-        # hub-table pool is phase-static: allocate once, reset per iteration
-        pk = (inp.rowptr.data_ptr(), n_pool, inp.weights.dtype)
-        hit = _pool_cache.get(pk)
-        if hit is None:
-            if len(_pool_cache) > 4:
-                _pool_cache.clear()
-            pool_keys = torch.empty(n_pool, dtype=torch.int32, device=dev)
-            pool_vals = torch.empty(n_pool, dtype=inp.weights.dtype,
-                                    device=dev)
-            _pool_cache[pk] = (pool_keys, pool_vals)
-        else:
-            pool_keys, pool_vals = hit
-        if n_pool:
-            pool_keys.fill_(-1)
-            pool_vals.zero_()
     if os.environ.get("CUVITE_PROGRESS"):
         import sys
         import time
         torch.cuda.synchronize()
         sizes = [int(v.numel()) for v in vlists]
-        print(f"[move] classes {sizes} pool={n_pool}", file=sys.stderr,
-              flush=True)
+        print(f"[move] classes {sizes} hubs={hubs64.numel()}",
+              file=sys.stderr, flush=True)
         outs = []
-        for i in range(6):
+        for i in range(5):
             one = [v if j == i else v[:0] for j, v in enumerate(vlists)]
             t0 = time.perf_counter()
             outs.append(ext.local_move_bucketed(
                 inp.rowptr, inp.tails, inp.weights, inp.curr_comm,
                 inp.v_degree, inp.comm_size, inp.comm_degree, inp.comm_gid,
-                float(inp.constant), one, offsets, eoffs,
-                n_hub_edges if i == 5 else 0, pool_keys, pool_vals))
+                float(inp.constant), one))
             torch.cuda.synchronize()
             print(f"[move] class {i} n={sizes[i]} "
                   f"{time.perf_counter() - t0:.3f}s", file=sys.stderr,
@@ -349,14 +312,13 @@ def local_move(inp):
         # merge: each class wrote its own vertices; take per-class targets
         target = outs[0][0]
         cw = outs[0][1]
-        for i in range(1, 6):
+        for i in range(1, 5):
             vl = vlists[i].to(torch.int64)
             if vl.numel():
                 target[vl] = outs[i][0][vl]
                 cw[vl] = outs[i][1][vl]
-        if not hub_hip and hubs64.numel():
+        if hubs64.numel():
             t0 = time.perf_counter()
-            hdeg = eoffs[1:] - eoffs[:-1]
             hub_tgt, hub_cw = _hub_moves_sorted(inp, hubs64, hdeg)
             target[hubs64] = hub_tgt
             cw[hubs64] = hub_cw
@@ -365,13 +327,13 @@ def local_move(inp):
                   f"{time.perf_counter() - t0:.3f}s", file=sys.stderr,
                   flush=True)
         return target, cw
-    run_hub_sort = not hub_hip and hubs64.numel() > 0
+    run_hub_sort = hubs64.numel() > 0
     overlap = run_hub_sort and not os.environ.get("CUVITE_NO_OVERLAP")
     if overlap:
         # disjoint vertex sets: run the HIP class kernels on a side stream
-        # concurrently with the (longer) hub sort on the main stream; hub
-        # results come back as separate tensors and are scattered after the
-        # join (no cross-stream writes into the class kernels' outputs)
+        # concurrently with the (longer) hub pipeline on the main stream;
+        # hub results come back as separate tensors and are scattered after
+        # the join (no cross-stream writes into the class kernels' outputs)
         main = torch.cuda.current_stream(dev)
         side = _side_stream(dev)
         side.wait_stream(main)
@@ -379,9 +341,7 @@ def local_move(inp):
             target, cw = ext.local_move_bucketed(
                 inp.rowptr, inp.tails, inp.weights, inp.curr_comm,
                 inp.v_degree, inp.comm_size, inp.comm_degree, inp.comm_gid,
-                float(inp.constant), vlists, offsets, eoffs, n_hub_edges,
-                pool_keys, pool_vals)
-        hdeg = eoffs[1:] - eoffs[:-1]
+                float(inp.constant), vlists)
         hub_tgt, hub_cw = _hub_moves_sorted(inp, hubs64, hdeg)
         main.wait_stream(side)
         target.record_stream(main)
@@ -392,9 +352,8 @@ def local_move(inp):
     target, cw = ext.local_move_bucketed(
         inp.rowptr, inp.tails, inp.weights, inp.curr_comm, inp.v_degree,
         inp.comm_size, inp.comm_degree, inp.comm_gid, float(inp.constant),
-        vlists, offsets, eoffs, n_hub_edges, pool_keys, pool_vals)
+        vlists)
     if run_hub_sort:
-        hdeg = eoffs[1:] - eoffs[:-1]
         hub_tgt, hub_cw = _hub_moves_sorted(inp, hubs64, hdeg)
         target[hubs64] = hub_tgt
         cw[hubs64] = hub_cw

@@ -29,11 +29,6 @@ void launch_sub(const int32_t*, int, const MoveArgs<W>&, hipStream_t);
 template <typename W, int CAP>
 void launch_block(const int32_t*, int, const MoveArgs<W>&, hipStream_t);
 template <typename W>
-void launch_hub(const int32_t*, int, const int64_t*, int64_t, const int64_t*,
-                int32_t*, W*, double*, int32_t*, double*, int64_t*, int32_t*,
-                const MoveArgs<W>&, hipStream_t);
-constexpr int HUB_SPLITS_HOST = 16;  // keep in sync with HUB_SPLITS
-template <typename W>
 void launch_modularity(const W*, const W*, int64_t, double*, hipStream_t);
 template <typename W>
 void launch_scatter_add(W*, const int64_t*, const W*, int64_t, hipStream_t);
@@ -91,23 +86,22 @@ cuvite::MoveArgs<W> make_args(const at::Tensor& rowptr, const at::Tensor& tails,
       cw.data_ptr<W>()};
 }
 
-// vlists: [class0, class1, class2, block_class, global_class] vertex lists
+// vlists: the 5 LDS degree-class vertex lists (hubs excluded)
 // (int32, padded is NOT required; launchers pad logically by bounds checks in
 // the sub kernels via nlist).
 std::vector<at::Tensor> local_move(
     at::Tensor rowptr, at::Tensor tails, at::Tensor weights,
     at::Tensor curr_comm, at::Tensor v_degree, at::Tensor comm_size,
     at::Tensor comm_degree, at::Tensor comm_gid, double constant,
-    std::vector<at::Tensor> vlists, at::Tensor global_offsets,
-    at::Tensor hub_eoffs, int64_t total_hub_edges, at::Tensor pool_keys,
-    at::Tensor pool_vals) {
+    std::vector<at::Tensor> vlists) {
   CHECK_DEV(rowptr); CHECK_CONT(rowptr);
   CHECK_DEV(tails); CHECK_CONT(tails);
   CHECK_DEV(weights); CHECK_CONT(weights);
   CHECK_DEV(curr_comm); CHECK_CONT(curr_comm);
   TORCH_CHECK(tails.scalar_type() == at::kInt, "tails must be int32");
   TORCH_CHECK(curr_comm.scalar_type() == at::kInt, "curr_comm must be int32");
-  TORCH_CHECK(vlists.size() == 6, "expected 6 degree-class vertex lists");
+  TORCH_CHECK(vlists.size() == 5, "expected 5 degree-class vertex lists"
+              " (hub vertices go through the hub_moves pipeline)");
 
   const int64_t nv = rowptr.numel() - 1;
   auto target = curr_comm.narrow(0, 0, nv).clone();
@@ -134,27 +128,6 @@ std::vector<at::Tensor> local_move(
     if (vlists[4].numel())
       cuvite::launch_block<W, 8192>(vlists[4].data_ptr<int32_t>(),
                                     (int)vlists[4].numel(), args, stream);
-    if (vlists[5].numel()) {
-      const int nhub = (int)vlists[5].numel();
-      auto hub_self = at::zeros({nhub}, rowptr.options().dtype(at::kDouble));
-      auto overflow = at::zeros({1}, rowptr.options().dtype(at::kInt));
-      auto p_gain = at::empty({nhub * cuvite::HUB_SPLITS_HOST},
-                              rowptr.options().dtype(at::kDouble));
-      auto p_gid = at::empty({nhub * cuvite::HUB_SPLITS_HOST},
-                             rowptr.options().dtype(at::kLong));
-      auto p_dense = at::empty({nhub * cuvite::HUB_SPLITS_HOST},
-                               rowptr.options().dtype(at::kInt));
-      cuvite::launch_hub<W>(vlists[5].data_ptr<int32_t>(), nhub,
-                            hub_eoffs.data_ptr<int64_t>(), total_hub_edges,
-                            global_offsets.data_ptr<int64_t>(),
-                            pool_keys.data_ptr<int32_t>(),
-                            pool_vals.data_ptr<W>(),
-                            hub_self.data_ptr<double>(),
-                            overflow.data_ptr<int32_t>(),
-                            p_gain.data_ptr<double>(),
-                            p_gid.data_ptr<int64_t>(),
-                            p_dense.data_ptr<int32_t>(), args, stream);
-    }
   });
   C10_HIP_CHECK(hipGetLastError());
   return {target, cw};
@@ -221,76 +194,6 @@ std::vector<at::Tensor> csr_from_edges(int64_t nv, int64_t base,
   });
   C10_HIP_CHECK(hipGetLastError());
   return {rowptr, tails, weights};
-}
-
-std::vector<at::Tensor> hub_candidates(at::Tensor tails_flat,
-                                       at::Tensor weights_flat,
-                                       at::Tensor seg_flat,
-                                       at::Tensor curr_comm, at::Tensor eoffs,
-                                       int64_t C) {
-  CHECK_DEV(tails_flat); CHECK_CONT(tails_flat);
-  CHECK_DEV(weights_flat); CHECK_CONT(weights_flat);
-  CHECK_DEV(seg_flat); CHECK_CONT(seg_flat);
-  CHECK_DEV(curr_comm); CHECK_CONT(curr_comm);
-  CHECK_DEV(eoffs); CHECK_CONT(eoffs);
-  TORCH_CHECK(tails_flat.scalar_type() == at::kInt);
-  TORCH_CHECK(seg_flat.scalar_type() == at::kInt);
-  const int64_t n = tails_flat.numel();
-  // reduce_by_key writes its unique count as unsigned int; the argmax kernel
-  // reads it through an int32 pointer — enforce the range here instead of
-  // relying on the Python-side chunk cap
-  TORCH_CHECK(n < (int64_t)INT32_MAX,
-              "hub candidate batch exceeds int32 count range; chunk the hubs");
-  const int nseg = (int)(eoffs.numel() - 1);
-  auto stream = at::hip::getCurrentHIPStream().stream();
-  int end_bit = 1;
-  while ((int64_t(1) << end_bit) < C) end_bit++;
-  auto uniq = at::empty({n}, eoffs.options());
-  auto cnt = at::zeros({1}, tails_flat.options());  // int32 count
-  auto sums = at::empty({n}, weights_flat.options());
-  AT_DISPATCH_FLOATING_TYPES(weights_flat.scalar_type(), "hub_cand", [&] {
-    using W = scalar_t;
-    auto keys = at::empty({n}, tails_flat.options());
-    cuvite::launch_gather_comm(tails_flat.data_ptr<int32_t>(),
-                               curr_comm.data_ptr<int32_t>(), n,
-                               keys.data_ptr<int32_t>(), stream);
-    auto keys2 = at::empty({n}, tails_flat.options());
-    auto vals2 = at::empty({n}, weights_flat.options());
-    size_t bytes = 0;
-    cuvite::segsort_pairs<W>(nullptr, &bytes, keys.data_ptr<int32_t>(),
-                             keys2.data_ptr<int32_t>(),
-                             weights_flat.data_ptr<W>(),
-                             vals2.data_ptr<W>(), n, nseg,
-                             eoffs.data_ptr<int64_t>(), end_bit, stream);
-    auto temp = at::empty({(int64_t)bytes},
-                          tails_flat.options().dtype(at::kByte));
-    cuvite::segsort_pairs<W>(temp.data_ptr(), &bytes,
-                             keys.data_ptr<int32_t>(),
-                             keys2.data_ptr<int32_t>(),
-                             weights_flat.data_ptr<W>(),
-                             vals2.data_ptr<W>(), n, nseg,
-                             eoffs.data_ptr<int64_t>(), end_bit, stream);
-    auto key64 = at::empty({n}, eoffs.options());
-    cuvite::launch_pack_key64(keys2.data_ptr<int32_t>(),
-                              seg_flat.data_ptr<int32_t>(), n, C,
-                              key64.data_ptr<int64_t>(), stream);
-    size_t bytes2 = 0;
-    cuvite::reduce_by_key64<W>(nullptr, &bytes2, key64.data_ptr<int64_t>(),
-                               vals2.data_ptr<W>(), n,
-                               uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
-                               (unsigned int*)cnt.data_ptr<int32_t>(),
-                               stream);
-    auto temp2 = at::empty({(int64_t)bytes2},
-                           tails_flat.options().dtype(at::kByte));
-    cuvite::reduce_by_key64<W>(temp2.data_ptr(), &bytes2,
-                               key64.data_ptr<int64_t>(),
-                               vals2.data_ptr<W>(), n,
-                               uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
-                               (unsigned int*)cnt.data_ptr<int32_t>(),
-                               stream);
-  });
-  C10_HIP_CHECK(hipGetLastError());
-  return {uniq, sums, cnt};
 }
 
 // Full hub move: gather -> segmented sort -> reduce_by_key -> wave-per-hub
@@ -431,8 +334,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_sum", &row_sum, "per-row CSR weight sum (HIP)");
   m.def("apply_deltas_", &apply_deltas_,
         "fused community size/degree delta update for moved vertices (HIP)");
-  m.def("hub_candidates", &hub_candidates,
-        "segmented-sort + reduce_by_key hub candidate generation (rocPRIM)");
   m.def("hub_moves", &hub_moves,
         "full device-side hub move: segsort + reduce_by_key + argmax");
 }

@@ -17,17 +17,19 @@
 //                                (48 KB -> 3 blocks/CU; the 8192-slot table
 //                                is 96 KB and caps residency at 1 block/CU)
 //   block B: deg in (2048, 4096] one 256-thread block/vertex, 8192-slot LDS
-//   hub    : deg > 4096          DEFAULT: hub_moves binding (rocPRIM
-//                                narrow-bit segmented radix sort +
-//                                reduce_by_key + wave-per-hub argmax) —
-//                                2.9x faster than the torch global-sort
-//                                fallback at s26 (A/B in profiles/;
-//                                CUVITE_HUB_SEGSORT=0 restores the
-//                                fallback). The hash-table pipeline below
-//                                is kept behind CUVITE_HUB_HIP (global
+//   hub    : deg > 4096          hub_moves binding (rocPRIM narrow-bit
+//                                segmented radix sort + reduce_by_key +
+//                                split-block argmax) — 2.9x faster than
+//                                the torch global-sort fallback at s26
+//                                (A/B in profiles/; CUVITE_HUB_SEGSORT=0
+//                                restores the fallback). A global-memory
+//                                hash-table hub pipeline existed in round
+//                                1 but was DELETED: open-addressing CAS
 //                                tables past the 4 MB XCD L2 measured
-//                                pathologically slow;
-//                                profiles/hub_pathology_and_s26.md)
+//                                pathologically slow with an unexplained
+//                                hang at exactly 2^21 slots
+//                                (profiles/hub_pathology_and_s26.md), and
+//                                the sort pipeline beats it anyway.
 //
 // All gain arithmetic is fp64 regardless of the weight dtype so trajectories
 // match the fp64 CPU oracle (tie-break on equal gains -> smaller GLOBAL id).
@@ -292,185 +294,6 @@ __global__ __launch_bounds__(BLOCK) void lv_move_block(
       tgt = cc;
     target[v] = tgt;
     cluster_weight[v] = wcc;
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Hub pipeline (deg > 4096): three edge/slot-parallel kernels instead of the
-// reference's one-block-per-vertex histogram (distGetMaxIndex_large_new,
-// louvain_cuda.cu:878-1022). A single R-MAT mega-hub (degree in the
-// millions) must not pin one CU: inserts are grid-strided over ALL hub
-// edges (binary search locates the owning hub), the argmax scan is split
-// into HUB_SPLITS blocks per hub with a small partial-reduction array, and
-// a finalize wave combines partials. Tables live in a global pool sized
-// 2*deg rounded to a power of two per hub (288 GB HBM: sized from the
-// graph, unlike the reference's hard-coded 2x2.7 GB buffers,
-// GpuGraph.cu:61-64). Every probe loop is bounded by cap as a hang guard
-// (sets `overflow` instead of spinning). Overflow is impossible by
-// construction: caps are 2*(deg+1) rounded up to a power of two
-// (ops/__init__.py _buckets_for), so distinct keys <= deg < cap/2 and a
-// linear probe always finds an empty slot; the flag exists solely so a
-// memory-corruption bug degrades to a wrong answer instead of a GPU hang,
-// which is why the host discards it on the (env-gated, experimental)
-// CUVITE_HUB_HIP path.
-// ---------------------------------------------------------------------------
-
-constexpr int HUB_SPLITS = 16;
-
-template <typename W, int BLOCK>
-__global__ __launch_bounds__(BLOCK) void hub_insert_kernel(
-    const int32_t* __restrict__ vlist, int nlist,
-    const int64_t* __restrict__ toffs,  // [nlist+1] table offsets
-    int32_t* __restrict__ pool_keys, W* __restrict__ pool_vals,
-    const int64_t* __restrict__ rowptr, const int32_t* __restrict__ tails,
-    const W* __restrict__ weights, const int32_t* __restrict__ curr_comm,
-    double* __restrict__ hub_self, int32_t* __restrict__ overflow) {
-  // HUB_SPLITS blocks per hub, each inserting a contiguous slice of the
-  // hub's adjacency: bounds the number of blocks contending on one table
-  // (a full-grid stride sprays atomics from every XCD onto one table and
-  // measured ~40x slower on a degree-2^19 hub)
-  constexpr int WAVES = BLOCK / 64;
-  __shared__ double red_self[WAVES];
-  const int hidx = blockIdx.x / HUB_SPLITS;
-  const int part = blockIdx.x % HUB_SPLITS;
-  const int32_t v = vlist[hidx];
-  const int64_t e0 = rowptr[v], e1 = rowptr[v + 1];
-  const int64_t deg = e1 - e0;
-  const int64_t s0 = e0 + part * deg / HUB_SPLITS;
-  const int64_t s1 = e0 + (part + 1) * deg / HUB_SPLITS;
-  const int64_t toff = toffs[hidx];
-  const int cap = (int)(toffs[hidx + 1] - toff);
-  int32_t* keys = pool_keys + toff;
-  W* vals = pool_vals + toff;
-  double selfloop = 0.0;
-  for (int64_t e = s0 + threadIdx.x; e < s1; e += BLOCK) {
-    const int32_t t = tails[e];
-    const W w = weights[e];
-    if (t == v) selfloop += (double)w;
-    const int32_t k = curr_comm[t];
-    uint32_t h = hash_u32((uint32_t)k) & (cap - 1);
-    for (int probes = 0; ; ++probes) {
-      const int32_t old = atomicCAS((int*)&keys[h], EMPTY_KEY, k);
-      if (old == EMPTY_KEY || old == k) {
-        unsafeAtomicAdd(&vals[h], w);
-        break;
-      }
-      h = (h + 1) & (cap - 1);
-      if (probes > cap) { atomicExch(overflow, 1); break; }
-    }
-  }
-  selfloop = sum_reduce<64>(selfloop);
-  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  if (lane == 0) red_self[wave] = selfloop;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    double tot = 0.0;
-#pragma unroll
-    for (int i = 0; i < WAVES; i++) tot += red_self[i];
-    if (tot != 0.0) unsafeAtomicAdd(&hub_self[hidx], tot);
-  }
-}
-
-// Guarded probe for the cluster-weight lookup (cc may be absent).
-template <typename W>
-DEV_INLINE W pool_probe(const int32_t* keys, const W* vals, int cap,
-                        int32_t k) {
-  uint32_t h = hash_u32((uint32_t)k) & (cap - 1);
-  for (int probes = 0; probes <= cap; ++probes) {
-    const int32_t kk = keys[h];
-    if (kk == k) return vals[h];
-    if (kk == EMPTY_KEY) return (W)0;
-    h = (h + 1) & (cap - 1);
-  }
-  return (W)0;
-}
-
-template <typename W, int BLOCK>
-__global__ __launch_bounds__(BLOCK) void hub_scan_kernel(
-    const int32_t* __restrict__ vlist, int nlist,
-    const int64_t* __restrict__ toffs, const int32_t* __restrict__ pool_keys,
-    const W* __restrict__ pool_vals, const int32_t* __restrict__ curr_comm,
-    const W* __restrict__ v_degree, const W* __restrict__ comm_degree,
-    const int64_t* __restrict__ comm_gid, const double* __restrict__ hub_self,
-    double constant, double* __restrict__ p_gain,
-    int64_t* __restrict__ p_gid, int32_t* __restrict__ p_dense) {
-  constexpr int WAVES = BLOCK / 64;
-  __shared__ double red_gain[WAVES];
-  __shared__ int64_t red_gid[WAVES];
-  __shared__ int32_t red_dense[WAVES];
-
-  const int hidx = blockIdx.x / HUB_SPLITS;
-  const int part = blockIdx.x % HUB_SPLITS;
-  const int32_t v = vlist[hidx];
-  const int32_t cc = curr_comm[v];
-  const int64_t toff = toffs[hidx];
-  const int64_t cap = toffs[hidx + 1] - toff;
-  const int32_t* keys = pool_keys + toff;
-  const W* vals = pool_vals + toff;
-
-  const W wcc = pool_probe(keys, vals, (int)cap, cc);
-  const double eix = (double)wcc - hub_self[hidx];
-  const double vdeg = (double)v_degree[v];
-  const double ax = (double)comm_degree[cc] - vdeg;
-
-  const int64_t s0 = part * cap / HUB_SPLITS;
-  const int64_t s1 = (part + 1) * cap / HUB_SPLITS;
-  Best best{0.0, comm_gid[cc], cc};
-  for (int64_t s = s0 + threadIdx.x; s < s1; s += BLOCK) {
-    const int32_t y = keys[s];
-    if (y == EMPTY_KEY || y == cc) continue;
-    const double eiy = (double)vals[s];
-    const double ay = (double)comm_degree[y];
-    const double g = 2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant;
-    best_combine(best, g, comm_gid[y], y);
-  }
-  best_reduce<64>(best);
-  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  if (lane == 0) {
-    red_gain[wave] = best.gain;
-    red_gid[wave] = best.gid;
-    red_dense[wave] = best.dense;
-  }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-#pragma unroll
-    for (int i = 1; i < WAVES; i++)
-      best_combine(best, red_gain[i], red_gid[i], red_dense[i]);
-    p_gain[blockIdx.x] = best.gain;
-    p_gid[blockIdx.x] = best.gid;
-    p_dense[blockIdx.x] = best.dense;
-  }
-}
-
-template <typename W, int BLOCK>
-__global__ __launch_bounds__(BLOCK) void hub_finalize_kernel(
-    const int32_t* __restrict__ vlist, int nlist,
-    const int64_t* __restrict__ toffs, const int32_t* __restrict__ pool_keys,
-    const W* __restrict__ pool_vals, const int32_t* __restrict__ curr_comm,
-    const int64_t* __restrict__ comm_size, const int64_t* __restrict__ comm_gid,
-    const double* __restrict__ p_gain, const int64_t* __restrict__ p_gid,
-    const int32_t* __restrict__ p_dense, int32_t* __restrict__ target,
-    W* __restrict__ cluster_weight) {
-  constexpr int WAVES = BLOCK / 64;
-  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-  const int hidx = blockIdx.x * WAVES + wave;
-  if (hidx >= nlist) return;
-  const int32_t v = vlist[hidx];
-  const int32_t cc = curr_comm[v];
-  Best best{0.0, comm_gid[cc], cc};
-  if (lane < HUB_SPLITS) {
-    const int p = hidx * HUB_SPLITS + lane;
-    best = Best{p_gain[p], p_gid[p], p_dense[p]};
-  }
-  best_reduce<64>(best);
-  if (lane == 0) {
-    int32_t tgt = best.dense;
-    if (comm_size[tgt] == 1 && comm_size[cc] == 1 && best.gid > comm_gid[cc])
-      tgt = cc;
-    target[v] = tgt;
-    cluster_weight[v] = pool_probe(pool_keys + toffs[hidx],
-                                   pool_vals + toffs[hidx],
-                                   (int)(toffs[hidx + 1] - toffs[hidx]), cc);
   }
 }
 
@@ -806,32 +629,6 @@ void launch_block(const int32_t* vlist, int nlist, const MoveArgs<W>& a,
                      a.constant, a.target, a.cluster_weight);
 }
 
-template <typename W>
-void launch_hub(const int32_t* vlist, int nlist, const int64_t* eoffs,
-                int64_t total_edges, const int64_t* toffs, int32_t* pool_keys,
-                W* pool_vals, double* hub_self, int32_t* overflow,
-                double* p_gain, int64_t* p_gid, int32_t* p_dense,
-                const MoveArgs<W>& a, hipStream_t stream) {
-  constexpr int BLOCK = 256;
-  (void)eoffs; (void)total_edges;
-  hipLaunchKernelGGL((hub_insert_kernel<W, BLOCK>),
-                     dim3(nlist * HUB_SPLITS), dim3(BLOCK), 0,
-                     stream, vlist, nlist, toffs, pool_keys, pool_vals,
-                     a.rowptr, a.tails, a.weights, a.curr_comm, hub_self,
-                     overflow);
-  hipLaunchKernelGGL((hub_scan_kernel<W, BLOCK>),
-                     dim3(nlist * HUB_SPLITS), dim3(BLOCK), 0, stream, vlist,
-                     nlist, toffs, pool_keys, pool_vals, a.curr_comm,
-                     a.v_degree, a.comm_degree, a.comm_gid, hub_self,
-                     a.constant, p_gain, p_gid, p_dense);
-  constexpr int WAVES = BLOCK / 64;
-  hipLaunchKernelGGL((hub_finalize_kernel<W, BLOCK>),
-                     dim3((nlist + WAVES - 1) / WAVES), dim3(BLOCK), 0,
-                     stream, vlist, nlist, toffs, pool_keys, pool_vals,
-                     a.curr_comm, a.comm_size, a.comm_gid, p_gain, p_gid,
-                     p_dense, a.target, a.cluster_weight);
-}
-
 // explicit instantiations used by bindings.cpp
 #define INSTANTIATE(W)                                                        \
   template void launch_sub<W, 16, 32>(const int32_t*, int, const MoveArgs<W>&,\
@@ -843,11 +640,7 @@ void launch_hub(const int32_t* vlist, int nlist, const int64_t* eoffs,
   template void launch_block<W, 4096>(const int32_t*, int,                  \
                                       const MoveArgs<W>&, hipStream_t);      \
   template void launch_block<W, 8192>(const int32_t*, int,                  \
-                                      const MoveArgs<W>&, hipStream_t);      \
-  template void launch_hub<W>(const int32_t*, int, const int64_t*, int64_t, \
-                              const int64_t*, int32_t*, W*, double*,         \
-                              int32_t*, double*, int64_t*, int32_t*,         \
-                              const MoveArgs<W>&, hipStream_t);
+                                      const MoveArgs<W>&, hipStream_t);
 
 INSTANTIATE(float)
 INSTANTIATE(double)
