@@ -33,8 +33,9 @@ def _require():
     return _ext
 
 
-# Degree-class boundaries (see louvain_kernels.hip header comment).
-_CLASS_BOUNDS = (16, 64, 512, 2048, 4096)
+# Degree-class boundaries (see louvain_kernels.hip header comment): sized
+# so the bulk classes' LDS tables stay at 24 KB/block (6 blocks/CU).
+_CLASS_BOUNDS = (16, 64, 256, 512, 1024, 2048, 4096)
 _bucket_cache: dict = {}
 
 
@@ -47,15 +48,13 @@ def _buckets_for(rowptr: torch.Tensor):
     if hit is not None and hit[0] is rowptr:  # identity check: ptr reuse safe
         return hit[1], hit[2], hit[3]
     deg = rowptr[1:] - rowptr[:-1]
-    b0, b1, b2, b3, b4 = _CLASS_BOUNDS
-    vlists = [
-        ((deg > 0) & (deg <= b0)).nonzero(as_tuple=True)[0].to(torch.int32),
-        ((deg > b0) & (deg <= b1)).nonzero(as_tuple=True)[0].to(torch.int32),
-        ((deg > b1) & (deg <= b2)).nonzero(as_tuple=True)[0].to(torch.int32),
-        ((deg > b2) & (deg <= b3)).nonzero(as_tuple=True)[0].to(torch.int32),
-        ((deg > b3) & (deg <= b4)).nonzero(as_tuple=True)[0].to(torch.int32),
-    ]
-    hubs64 = (deg > b4).nonzero(as_tuple=True)[0]
+    lo = 0
+    vlists = []
+    for b in _CLASS_BOUNDS:
+        vlists.append(((deg > lo) & (deg <= b)).nonzero(
+            as_tuple=True)[0].to(torch.int32))
+        lo = b
+    hubs64 = (deg > _CLASS_BOUNDS[-1]).nonzero(as_tuple=True)[0]
     hdeg = deg[hubs64]
     if len(_bucket_cache) > 8:
         _bucket_cache.clear()
@@ -298,7 +297,7 @@ def local_move(inp):
         print(f"[move] classes {sizes} hubs={hubs64.numel()}",
               file=sys.stderr, flush=True)
         outs = []
-        for i in range(5):
+        for i in range(len(vlists)):
             one = [v if j == i else v[:0] for j, v in enumerate(vlists)]
             t0 = time.perf_counter()
             outs.append(ext.local_move_bucketed(
@@ -312,7 +311,7 @@ def local_move(inp):
         # merge: each class wrote its own vertices; take per-class targets
         target = outs[0][0]
         cw = outs[0][1]
-        for i in range(1, 5):
+        for i in range(1, len(vlists)):
             vl = vlists[i].to(torch.int64)
             if vl.numel():
                 target[vl] = outs[i][0][vl]

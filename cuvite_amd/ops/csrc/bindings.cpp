@@ -100,7 +100,7 @@ std::vector<at::Tensor> local_move(
   CHECK_DEV(curr_comm); CHECK_CONT(curr_comm);
   TORCH_CHECK(tails.scalar_type() == at::kInt, "tails must be int32");
   TORCH_CHECK(curr_comm.scalar_type() == at::kInt, "curr_comm must be int32");
-  TORCH_CHECK(vlists.size() == 5, "expected 5 degree-class vertex lists"
+  TORCH_CHECK(vlists.size() == 7, "expected 7 degree-class vertex lists"
               " (hub vertices go through the hub_moves pipeline)");
 
   const int64_t nv = rowptr.numel() - 1;
@@ -120,14 +120,20 @@ std::vector<at::Tensor> local_move(
       cuvite::launch_sub<W, 64, 128>(vlists[1].data_ptr<int32_t>(),
                                      (int)vlists[1].numel(), args, stream);
     if (vlists[2].numel())
-      cuvite::launch_sub<W, 64, 1024>(vlists[2].data_ptr<int32_t>(),
-                                      (int)vlists[2].numel(), args, stream);
+      cuvite::launch_sub<W, 64, 512>(vlists[2].data_ptr<int32_t>(),
+                                     (int)vlists[2].numel(), args, stream);
     if (vlists[3].numel())
-      cuvite::launch_block<W, 4096>(vlists[3].data_ptr<int32_t>(),
-                                    (int)vlists[3].numel(), args, stream);
+      cuvite::launch_sub<W, 64, 1024>(vlists[3].data_ptr<int32_t>(),
+                                      (int)vlists[3].numel(), args, stream);
     if (vlists[4].numel())
-      cuvite::launch_block<W, 8192>(vlists[4].data_ptr<int32_t>(),
+      cuvite::launch_block<W, 2048>(vlists[4].data_ptr<int32_t>(),
                                     (int)vlists[4].numel(), args, stream);
+    if (vlists[5].numel())
+      cuvite::launch_block<W, 4096>(vlists[5].data_ptr<int32_t>(),
+                                    (int)vlists[5].numel(), args, stream);
+    if (vlists[6].numel())
+      cuvite::launch_block<W, 8192>(vlists[6].data_ptr<int32_t>(),
+                                    (int)vlists[6].numel(), args, stream);
   });
   C10_HIP_CHECK(hipGetLastError());
   return {target, cw};

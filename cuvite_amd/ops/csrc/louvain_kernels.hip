@@ -9,14 +9,18 @@
 // cooperative inserts, then computes the dQ argmax by scanning the table.
 //
 // Degree-class routing (the reference's 3-bucket idea, count_size_clmap
-// louvain_cuda.cu:1426-1592, rebuilt for wave64):
-//   class 0: deg in [1, 16]      16 lanes/vertex, 32-slot LDS table
+// louvain_cuda.cu:1426-1592, rebuilt for wave64). Class boundaries are
+// sized so the LDS table per block stays at <= 24 KB for the bulk classes
+// (6 resident blocks/CU on 160 KB LDS — the round-1 two-class split left
+// class (512,2048] on a 48 KB table at 3 blocks/CU, and PMC showed the
+// kernels latency-bound at 80% SQ_WAIT_ANY, so residency is the lever):
+//   class 0: deg in [1, 16]      16 lanes/vertex,  32-slot LDS table
 //   class 1: deg in (16, 64]     one wave/vertex, 128-slot LDS table
-//   class 2: deg in (64, 512]    one wave/vertex, 1024-slot LDS table
-//   block A: deg in (512, 2048]  one 256-thread block/vertex, 4096-slot LDS
-//                                (48 KB -> 3 blocks/CU; the 8192-slot table
-//                                is 96 KB and caps residency at 1 block/CU)
-//   block B: deg in (2048, 4096] one 256-thread block/vertex, 8192-slot LDS
+//   class 2: deg in (64, 256]    one wave/vertex, 512-slot LDS (24 KB/blk)
+//   class 3: deg in (256, 512]   one wave/vertex, 1024-slot LDS (48 KB/blk)
+//   class 4: deg in (512, 1024]  256-thread block/vertex, 2048-slot (24 KB)
+//   class 5: deg in (1024, 2048] 256-thread block/vertex, 4096-slot (48 KB)
+//   class 6: deg in (2048, 4096] 256-thread block/vertex, 8192-slot (96 KB)
 //   hub    : deg > 4096          hub_moves binding (rocPRIM narrow-bit
 //                                segmented radix sort + reduce_by_key +
 //                                split-block argmax) — 2.9x faster than
@@ -120,6 +124,45 @@ DEV_INLINE void table_insert(int32_t* keys, W* vals, int cap, int32_t k, W w) {
   }
 }
 
+// Edge-insert loop with a 4-deep software pipeline: the per-edge chain
+// (load tail -> gather curr_comm[tail] -> LDS CAS insert) is latency-bound
+// (PMC: ~80% SQ_WAIT_ANY at 3 blocks/CU); issuing 4 independent tail loads,
+// then 4 independent label gathers, before any insert overlaps the two
+// global-load latencies instead of serializing them per edge.
+template <typename W, int STRIDE>
+DEV_INLINE double insert_edges_pipelined(
+    int64_t e0, int64_t e1, int lane, int32_t v,
+    const int32_t* __restrict__ tails, const W* __restrict__ weights,
+    const int32_t* __restrict__ curr_comm, int32_t* keys, W* vals, int cap) {
+  constexpr int PIPE = 4;
+  double selfloop = 0.0;
+  int64_t e = e0 + lane;
+  for (; e + (PIPE - 1) * (int64_t)STRIDE < e1; e += PIPE * (int64_t)STRIDE) {
+    int32_t t[PIPE];
+    W w[PIPE];
+#pragma unroll
+    for (int k = 0; k < PIPE; k++) {
+      t[k] = tails[e + k * STRIDE];
+      w[k] = weights[e + k * STRIDE];
+    }
+    int32_t c[PIPE];
+#pragma unroll
+    for (int k = 0; k < PIPE; k++) c[k] = curr_comm[t[k]];
+#pragma unroll
+    for (int k = 0; k < PIPE; k++) {
+      if (t[k] == v) selfloop += (double)w[k];
+      table_insert(keys, vals, cap, c[k], w[k]);
+    }
+  }
+  for (; e < e1; e += STRIDE) {
+    int32_t t = tails[e];
+    W w = weights[e];
+    if (t == v) selfloop += (double)w;
+    table_insert(keys, vals, cap, curr_comm[t], w);
+  }
+  return selfloop;
+}
+
 // Scan the table slots owned by this lane and fold the argmax.
 template <typename W>
 DEV_INLINE void scan_slots(const int32_t* keys, const W* vals, int cap,
@@ -190,13 +233,8 @@ __global__ __launch_bounds__(BLOCK) void lv_move_sub(
   }
   __syncthreads();
 
-  double selfloop = 0.0;
-  for (int64_t e = e0 + lane; e < e1; e += LANES) {
-    int32_t t = tails[e];
-    W w = weights[e];
-    if (t == v) selfloop += (double)w;
-    table_insert(gkeys, gvals, cap, curr_comm[t], w);
-  }
+  double selfloop = insert_edges_pipelined<W, LANES>(
+      e0, e1, lane, v, tails, weights, curr_comm, gkeys, gvals, cap);
   __syncthreads();
 
   selfloop = sum_reduce<LANES>(selfloop);
@@ -256,13 +294,8 @@ __global__ __launch_bounds__(BLOCK) void lv_move_block(
   __syncthreads();
 
   const int32_t cc = curr_comm[v];
-  double selfloop = 0.0;
-  for (int64_t e = e0 + tid; e < e1; e += BLOCK) {
-    int32_t t = tails[e];
-    W w = weights[e];
-    if (t == v) selfloop += (double)w;
-    table_insert(keys, vals, cap, curr_comm[t], w);
-  }
+  double selfloop = insert_edges_pipelined<W, BLOCK>(
+      e0, e1, tid, v, tails, weights, curr_comm, keys, vals, cap);
   selfloop = sum_reduce<64>(selfloop);
   if (lane == 0) red_self[wave] = selfloop;
   __syncthreads();
@@ -635,8 +668,12 @@ void launch_block(const int32_t* vlist, int nlist, const MoveArgs<W>& a,
                                       hipStream_t);                           \
   template void launch_sub<W, 64, 128>(const int32_t*, int,                  \
                                        const MoveArgs<W>&, hipStream_t);     \
+  template void launch_sub<W, 64, 512>(const int32_t*, int,                  \
+                                       const MoveArgs<W>&, hipStream_t);     \
   template void launch_sub<W, 64, 1024>(const int32_t*, int,                 \
                                         const MoveArgs<W>&, hipStream_t);    \
+  template void launch_block<W, 2048>(const int32_t*, int,                  \
+                                      const MoveArgs<W>&, hipStream_t);      \
   template void launch_block<W, 4096>(const int32_t*, int,                  \
                                       const MoveArgs<W>&, hipStream_t);      \
   template void launch_block<W, 8192>(const int32_t*, int,                  \
