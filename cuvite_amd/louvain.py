@@ -581,6 +581,9 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
             if cfg.one_phase:
                 break
             t0 = time.perf_counter()
+            if dev.type == "cuda":
+                from . import ops
+                ops.clear_phase_caches()  # free phase-keyed GPU tensors
             level, renum = coarsen(level, comm, cvect, halo=phase_halo)
             # cvect-composed orig_assign holds OLD comm gids; renumber them
             orig_assign = renum(orig_assign)

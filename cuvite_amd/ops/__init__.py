@@ -65,6 +65,15 @@ def _buckets_for(rowptr: torch.Tensor):
 _side_streams: dict = {}
 
 
+def clear_phase_caches():
+    """Drop the phase-keyed caches (degree buckets, deduped hub adjacency,
+    hub chunk groups). Called by the phase loop before coarsening: at s27
+    the cached hub arrays are tens of GB and the next level re-keys anyway."""
+    _bucket_cache.clear()
+    _hub_static_cache.clear()
+    _hub_groups_cache.clear()
+
+
 def _side_stream(dev):
     st = _side_streams.get(dev)
     if st is None:
@@ -185,12 +194,15 @@ def _hub_static(inp, hubs, hdeg):
         selfloop.index_add_(0, seg[selfmask], wts[selfmask])
     del selfmask
     if _hub_segsort_enabled() and dev.type == "cuda":
-        # int32 copies + deduped per-hub offsets for the rocPRIM path
+        # int32 copies + deduped per-hub offsets for the rocPRIM path;
+        # the int64 originals are not needed again on this path — drop them
+        # (at s27 they are ~24 B per deduped hub edge, tens of GB)
         offs_d = torch.zeros(nhub + 1, dtype=torch.int64, device=dev)
         offs_d[1:] = torch.cumsum(
             torch.bincount(seg, minlength=nhub), dim=0)
         extra = (tails_h.to(torch.int32), seg.to(torch.int32), offs_d,
                  wts.to(inp.weights.dtype), hubs.to(torch.int32))
+        seg = tails_h = wts = None
     else:
         extra = None
     data = (seg, tails_h, wts, selfloop, extra)
