@@ -24,8 +24,11 @@ import time
 
 # expandable segments kill the fragmentation that OOM'd the s27 converged
 # run at 254 GB allocated + 24 GB reserved-but-unallocated; must be set
-# before torch initializes its allocator
-os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+# before torch initializes its allocator (all naming generations, ROCm
+# reads the CUDA/HIP aliases)
+for _k in ("PYTORCH_ALLOC_CONF", "PYTORCH_CUDA_ALLOC_CONF",
+           "PYTORCH_HIP_ALLOC_CONF"):
+    os.environ.setdefault(_k, "expandable_segments:True")
 
 import torch  # noqa: E402
 
@@ -129,9 +132,18 @@ def main():
     if do_conv:
         from cuvite_amd.louvain import louvain
         _p("converged multi-phase run starting")
+        # reuse the timed region's halo for phase 0 and drop tensors the
+        # converged run can no longer reach: the int64 CSR tails are dead
+        # once tails_dense exists (s27: 34 GB) and the timed PhaseState's
+        # aggregates are superseded
+        phase0_halo = state.halo
+        if device.type == "cuda" and comm.world == 1:
+            dg.g.tails = torch.empty(0, dtype=dg.g.tails.dtype,
+                                     device=device)
         t0 = time.perf_counter()
         try:
-            res = louvain(dg, comm, LouvainConfig(backend=args.backend))
+            res = louvain(dg, comm, LouvainConfig(backend=args.backend),
+                          halo=phase0_halo)
             if device.type == "cuda":
                 torch.cuda.synchronize()
             conv_s = time.perf_counter() - t0

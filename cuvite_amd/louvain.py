@@ -521,10 +521,13 @@ def _modularity(state: PhaseState) -> float:
 
 
 def louvain(dg: DistGraph, comm: Optional[Comm] = None,
-            cfg: Optional[LouvainConfig] = None) -> LouvainResult:
+            cfg: Optional[LouvainConfig] = None,
+            halo=None) -> LouvainResult:
     """Full multi-phase Louvain (ref main.cpp:218-495). Returns the final
     community id per ORIGINAL local vertex (contiguous global ids) and the
-    final modularity."""
+    final modularity. `halo`: optional prebuilt HaloContext for the INPUT
+    graph (phase 0) — callers that already built one (bench.py's timed
+    region) pass it to avoid a duplicate ghost structure (17 GB at s27)."""
     from .coarsen import coarsen, remap_labels
     from .coloring import distance1_coloring
 
@@ -549,11 +552,12 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
 
         colors = None
         num_colors = 0
-        pre_halo = None
+        pre_halo = halo if phase == 0 else None
         if (cfg.coloring or cfg.ordering) and phase == 0:
             from .halo import build_halo as _bh
             t0 = time.perf_counter()
-            pre_halo = _bh(level, comm)  # shared with PhaseState below
+            if pre_halo is None:
+                pre_halo = _bh(level, comm)  # shared with PhaseState below
             colors, num_colors = distance1_coloring(
                 level, comm, n_hash=max(1, cfg.max_colors // 2),
                 halo=pre_halo)
