@@ -143,3 +143,29 @@ def test_et_freeze_semantics():
         _et_update(state, cfg, tgt, rng)
         state.past_comm = state.curr_comm.clone()
     assert not bool(state.active.any())  # every vertex frozen after 3 stable
+
+
+def test_louvain_with_prebuilt_halo_matches():
+    """louvain(halo=...) (bench.py's converged-run path: reuse the timed
+    region's phase-0 halo) must produce the identical result."""
+    from cuvite_amd.halo import build_halo
+    g = rmat_graph(8, 8, seed=3)
+    dg = single_partition(g)
+    ref = louvain(dg, Comm(), LouvainConfig(backend="torch"))
+    h = build_halo(dg, Comm())
+    got = louvain(dg, Comm(), LouvainConfig(backend="torch"), halo=h)
+    assert got.modularity == ref.modularity
+    assert torch.equal(got.communities, ref.communities)
+    assert got.total_iters == ref.total_iters
+
+
+def test_louvain_with_prebuilt_halo_coloring():
+    """halo= composes with -c (coloring shares the provided halo)."""
+    from cuvite_amd.halo import build_halo
+    g = rmat_graph(8, 8, seed=3)
+    dg = single_partition(g)
+    cfg = LouvainConfig(backend="torch", coloring=True, max_colors=6)
+    ref = louvain(dg, Comm(), cfg)
+    got = louvain(dg, Comm(), cfg, halo=build_halo(dg, Comm()))
+    assert got.modularity == ref.modularity
+    assert torch.equal(got.communities, ref.communities)
