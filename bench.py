@@ -132,14 +132,11 @@ def main():
     if do_conv:
         from cuvite_amd.louvain import louvain
         _p("converged multi-phase run starting")
-        # reuse the timed region's halo for phase 0 and drop tensors the
-        # converged run can no longer reach: the int64 CSR tails are dead
-        # once tails_dense exists (s27: 34 GB) and the timed PhaseState's
-        # aggregates are superseded
+        # reuse the timed region's halo for phase 0 (avoids a duplicate
+        # tails_dense — 17 GB at s27). NOTE: do NOT free dg.g.tails here:
+        # Graph.ne derives from it and an emptied tails silently zeroes
+        # v_degree via the ne==0 short-circuit (caught on s27).
         phase0_halo = state.halo
-        if device.type == "cuda" and comm.world == 1:
-            dg.g.tails = torch.empty(0, dtype=dg.g.tails.dtype,
-                                     device=device)
         t0 = time.perf_counter()
         try:
             res = louvain(dg, comm, LouvainConfig(backend=args.backend),

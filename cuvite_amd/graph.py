@@ -206,6 +206,10 @@ class DistGraph:
     def local_degree_sum(self) -> torch.Tensor:
         """Per-vertex weighted degree (vDegree; ref louvain.cpp:2126-2151)."""
         nv = self.nv
+        # loud guard: an emptied/mismatched tails array would silently zero
+        # every degree here (one 8-byte D2H read, once per phase)
+        assert self.g.ne == int(self.g.rowptr[-1]), \
+            "CSR tails/rowptr inconsistent (tails freed or truncated?)"
         if self.g.ne == 0:
             return torch.zeros(nv, dtype=self.g.weights.dtype, device=self.g.device)
         if self.g.device.type == "cuda":
