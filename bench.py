@@ -132,15 +132,20 @@ def main():
     if do_conv:
         from cuvite_amd.louvain import louvain
         _p("converged multi-phase run starting")
-        # reuse the timed region's halo for phase 0 (avoids a duplicate
-        # tails_dense — 17 GB at s27). NOTE: do NOT free dg.g.tails here:
-        # Graph.ne derives from it and an emptied tails silently zeroes
-        # v_degree via the ne==0 short-circuit (caught on s27).
-        phase0_halo = state.halo
+        # memory handoff for the giant-graph case (s27: the converged run
+        # OOM'd twice before this): reuse the timed region's halo for phase
+        # 0 (no duplicate 17 GB tails_dense), drop the timed PhaseState, and
+        # release the int64 CSR tails (34 GB; ne stays valid via
+        # release_tails). The halo travels through a single-ref handoff so
+        # louvain() can drop it once phase 0's coarsening is done.
+        conv_halo = [state.halo]
+        del state
+        if device.type == "cuda" and comm.world == 1:
+            dg.g.release_tails()
         t0 = time.perf_counter()
         try:
             res = louvain(dg, comm, LouvainConfig(backend=args.backend),
-                          halo=phase0_halo)
+                          halo=conv_halo.pop())
             if device.type == "cuda":
                 torch.cuda.synchronize()
             conv_s = time.perf_counter() - t0

@@ -179,6 +179,21 @@ def _aggregate(s: torch.Tensor, t: torch.Tensor, w: torch.Tensor, gnc: int):
     """Merge duplicate (s, t) directed edges, summing weights; returns sorted
     by (s, t). Keys fit int64 because gnc < 2^31."""
     key = s * gnc + t
+    if key.is_cuda:
+        from . import ops
+        if ops.available():
+            # narrow-bit rocPRIM radix sort (only bits of gnc^2, ~50 at s26
+            # vs 64 for a generic int64 sort) + reduce_by_key — replaces the
+            # sort / gather / unique_consecutive / double-cumsum chain that
+            # dominated the 5.9 s phase-0 rebuild (profiles/NEXT.md)
+            end_bit = max(1, (gnc * gnc - 1).bit_length())
+            w64 = w if w.dtype == torch.float64 else w.to(torch.float64)
+            uniq, sums, cnt = ops._require().sort_reduce_pairs(
+                key, w64, min(end_bit, 64))
+            m = int(cnt[0])
+            uniq = uniq[:m]
+            sums = sums[:m]
+            return uniq // gnc, uniq % gnc, sums.to(w.dtype)
     key_s, order = torch.sort(key)
     w_s = w[order]
     uniq, counts = torch.unique_consecutive(key_s, return_counts=True)

@@ -96,6 +96,7 @@ class Graph:
         self.rowptr = rowptr
         self.tails = tails
         self.weights = weights
+        self._ne_released: int = -1  # set by release_tails()
 
     @property
     def nv(self) -> int:
@@ -103,7 +104,19 @@ class Graph:
 
     @property
     def ne(self) -> int:
+        if self._ne_released >= 0:
+            return self._ne_released
         return self.tails.numel()
+
+    def release_tails(self):
+        """Free the global-id tails array while keeping `ne` valid. Once a
+        HaloContext exists, every compute path reads its dense int32 tails
+        instead — the int64 originals are 34 GB of dead weight at R-MAT s27
+        and were the difference between the converged multi-phase run
+        fitting in 288 GB HBM or not. After release, ghost discovery /
+        fresh halo builds on this graph raise."""
+        self._ne_released = self.tails.numel()
+        self.tails = torch.empty(0, dtype=torch.int64, device=self.device)
 
     @property
     def device(self):
@@ -190,6 +203,9 @@ class DistGraph:
         """Sorted unique global ids of remote tails (the rank's ghosts).
         Chunked: torch advanced indexing overflows internal int32 offsets
         past ~2^31 elements (observed as an absurd-size alloc at R-MAT s26)."""
+        if self.g._ne_released >= 0:
+            raise RuntimeError("tails released: build halos before "
+                               "Graph.release_tails()")
         t = self.g.tails
         if self.partition.nranks == 1:
             return torch.empty(0, dtype=torch.int64, device=t.device)

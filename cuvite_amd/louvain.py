@@ -566,9 +566,11 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
         ne_global = comm.allreduce_scalar(float(level.ne)) \
             if comm.world > 1 else float(level.ne)
         t0 = time.perf_counter()
+        halo = None  # phase-0 halo ownership moves to run_phase/coarsen
         curr_mod, cvect, iters, phase_halo = run_phase(
             level, comm, cfg, curr_mod, threshold,
             colors=colors, num_colors=num_colors, halo=pre_halo)
+        pre_halo = None
         t_cluster = time.perf_counter() - t0
         times["clustering"] += t_cluster
         tot_iters += iters
@@ -589,6 +591,7 @@ def louvain(dg: DistGraph, comm: Optional[Comm] = None,
                 from . import ops
                 ops.clear_phase_caches()  # free phase-keyed GPU tensors
             level, renum = coarsen(level, comm, cvect, halo=phase_halo)
+            phase_halo = None  # free the ghost structure before next phase
             # cvect-composed orig_assign holds OLD comm gids; renumber them
             orig_assign = renum(orig_assign)
             level_entry["rebuild_s"] = time.perf_counter() - t0

@@ -51,6 +51,9 @@ template <typename W>
 void reduce_by_key64(void*, size_t*, const int64_t*, const W*, int64_t,
                      int64_t*, W*, unsigned int*, hipStream_t);
 template <typename W>
+void sort_pairs64(void*, size_t*, const int64_t*, int64_t*, const W*, W*,
+                  int64_t, int, hipStream_t);
+template <typename W>
 void launch_hub_argmax(const int64_t*, const W*, const int32_t*, int64_t,
                        const int32_t*, int, const double*, const int32_t*,
                        const W*, const int64_t*, const W*, const int64_t*,
@@ -285,6 +288,55 @@ std::vector<at::Tensor> hub_moves(
   });
   C10_HIP_CHECK(hipGetLastError());
   return {target_hub, cw_hub};
+}
+
+// Coarsening aggregate: sort packed (src,dst) keys with weights (narrow-bit
+// radix) and sum duplicate keys. Returns (uniq int64 [n], sums W [n],
+// count int32[1]) — caller trims to count (ref fill_newEdgesMap +
+// duplicate-edge merge, rebuild.cpp:244-279, 379-411, done with nested
+// std::maps on the host there).
+std::vector<at::Tensor> sort_reduce_pairs(at::Tensor key, at::Tensor val,
+                                          int64_t end_bit) {
+  CHECK_DEV(key); CHECK_CONT(key);
+  CHECK_DEV(val); CHECK_CONT(val);
+  TORCH_CHECK(key.scalar_type() == at::kLong, "key must be int64");
+  TORCH_CHECK(key.numel() == val.numel());
+  const int64_t n = key.numel();
+  TORCH_CHECK(n < (int64_t)INT32_MAX, "chunk exceeds int32 count range");
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto uniq = at::empty({std::max<int64_t>(n, 1)}, key.options());
+  auto sums = at::empty({std::max<int64_t>(n, 1)}, val.options());
+  auto cnt = at::zeros({1}, key.options().dtype(at::kInt));
+  if (n == 0) return {uniq.narrow(0, 0, 0), sums.narrow(0, 0, 0), cnt};
+  AT_DISPATCH_FLOATING_TYPES(val.scalar_type(), "sort_reduce_pairs", [&] {
+    using W = scalar_t;
+    auto keys2 = at::empty({n}, key.options());
+    auto vals2 = at::empty({n}, val.options());
+    size_t bytes = 0;
+    cuvite::sort_pairs64<W>(nullptr, &bytes, key.data_ptr<int64_t>(),
+                            keys2.data_ptr<int64_t>(), val.data_ptr<W>(),
+                            vals2.data_ptr<W>(), n, (int)end_bit, stream);
+    auto temp = at::empty({(int64_t)bytes}, key.options().dtype(at::kByte));
+    cuvite::sort_pairs64<W>(temp.data_ptr(), &bytes, key.data_ptr<int64_t>(),
+                            keys2.data_ptr<int64_t>(), val.data_ptr<W>(),
+                            vals2.data_ptr<W>(), n, (int)end_bit, stream);
+    size_t bytes2 = 0;
+    cuvite::reduce_by_key64<W>(nullptr, &bytes2, keys2.data_ptr<int64_t>(),
+                               vals2.data_ptr<W>(), n,
+                               uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
+                               (unsigned int*)cnt.data_ptr<int32_t>(),
+                               stream);
+    auto temp2 = at::empty({(int64_t)bytes2},
+                           key.options().dtype(at::kByte));
+    cuvite::reduce_by_key64<W>(temp2.data_ptr(), &bytes2,
+                               keys2.data_ptr<int64_t>(),
+                               vals2.data_ptr<W>(), n,
+                               uniq.data_ptr<int64_t>(), sums.data_ptr<W>(),
+                               (unsigned int*)cnt.data_ptr<int32_t>(),
+                               stream);
+  });
+  C10_HIP_CHECK(hipGetLastError());
+  return {uniq, sums, cnt};
 }
 
 void apply_deltas_(at::Tensor target, at::Tensor curr, at::Tensor v_degree,

@@ -40,6 +40,7 @@
 
 #include <hip/hip_runtime.h>
 #include <rocprim/device/device_segmented_radix_sort.hpp>
+#include <rocprim/device/device_radix_sort.hpp>
 #include <rocprim/device/device_reduce_by_key.hpp>
 
 #include <cstdint>
@@ -843,6 +844,24 @@ template void segsort_pairs<float>(void*, size_t*, const int32_t*, int32_t*,
 template void segsort_pairs<double>(void*, size_t*, const int32_t*, int32_t*,
                                     const double*, double*, int64_t, int,
                                     const int64_t*, int, hipStream_t);
+
+// Narrow-bit global radix sort of packed (src, dst) coarse-edge keys with
+// their weights (coarsening aggregate; end_bit = bits of gnc^2, ~50 at s26
+// vs the 64 bits a generic int64 sort pays).
+template <typename W>
+void sort_pairs64(void* temp, size_t* bytes, const int64_t* keys_in,
+                  int64_t* keys_out, const W* vals_in, W* vals_out,
+                  int64_t n, int end_bit, hipStream_t stream) {
+  (void)rocprim::radix_sort_pairs(temp, *bytes, keys_in, keys_out, vals_in,
+                                  vals_out, (size_t)n, 0u,
+                                  (unsigned)end_bit, stream);
+}
+template void sort_pairs64<float>(void*, size_t*, const int64_t*, int64_t*,
+                                  const float*, float*, int64_t, int,
+                                  hipStream_t);
+template void sort_pairs64<double>(void*, size_t*, const int64_t*, int64_t*,
+                                   const double*, double*, int64_t, int,
+                                   hipStream_t);
 
 template <typename W>
 void reduce_by_key64(void* temp, size_t* bytes, const int64_t* keys,

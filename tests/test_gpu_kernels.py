@@ -328,3 +328,38 @@ def test_apply_deltas_matches_torch():
     ops.apply_deltas_(target, curr, vdeg, base, bound, size, degree)
     assert torch.equal(size, size_ref)
     assert torch.allclose(degree, degree_ref)
+
+
+def test_coarsen_aggregate_rocprim_matches_cpu():
+    """GPU _aggregate (rocPRIM narrow-bit sort_reduce_pairs) vs the CPU
+    sort/cumsum reference on random duplicate-heavy coarse edges."""
+    from cuvite_amd.coarsen import _aggregate
+    torch.manual_seed(3)
+    n, gnc = 200000, 997
+    s = torch.randint(0, gnc, (n,), dtype=torch.int64)
+    t = torch.randint(0, gnc, (n,), dtype=torch.int64)
+    w = torch.rand(n, dtype=torch.float64)
+    cs, ct, cw = _aggregate(s, t, w, gnc)
+    dev = torch.device("cuda:0")
+    gs, gt, gw = _aggregate(s.to(dev), t.to(dev), w.to(dev), gnc)
+    assert torch.equal(gs.cpu(), cs)
+    assert torch.equal(gt.cpu(), ct)
+    assert torch.allclose(gw.cpu(), cw, atol=1e-12)
+
+
+def test_release_tails_louvain_still_runs():
+    """bench.py's converged-run memory handoff: after build_halo +
+    release_tails, louvain(halo=...) must produce the same result."""
+    from cuvite_amd.halo import build_halo
+    dev = torch.device("cuda:0")
+    from cuvite_amd.louvain import louvain, LouvainConfig
+    from cuvite_amd.parallel import Comm
+    g = rmat_graph(10, 16, seed=2).to(dev)
+    dg = single_partition(g)
+    ref = louvain(dg, Comm(dev), LouvainConfig(backend="hip"))
+    h = build_halo(dg, Comm(dev))
+    dg.g.release_tails()
+    assert dg.g.ne == int(dg.g.rowptr[-1])
+    got = louvain(dg, Comm(dev), LouvainConfig(backend="hip"), halo=h)
+    assert got.modularity == ref.modularity
+    assert torch.equal(got.communities, ref.communities)
