@@ -392,6 +392,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_sum", &row_sum, "per-row CSR weight sum (HIP)");
   m.def("apply_deltas_", &apply_deltas_,
         "fused community size/degree delta update for moved vertices (HIP)");
+  m.def("sort_reduce_pairs", &sort_reduce_pairs,
+        "narrow-bit radix sort + reduce_by_key coarse-edge aggregate "
+        "(rocPRIM)");
   m.def("hub_moves", &hub_moves,
         "full device-side hub move: segsort + reduce_by_key + argmax");
 }
