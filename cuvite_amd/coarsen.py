@@ -108,20 +108,8 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor,
     # fine edge list, so partial aggregation also bounds peak memory
     ne = dg.g.ne
     CH = 1 << 28
-    # chunk-aggregates are merged in one final sort while their total stays
-    # under the torch sort cap; past it (s30-per-rank territory) they are
-    # folded progressively so no single sort exceeds INT_MAX elements
-    MERGE_CAP = 3 << 29
     rowptr = dg.g.rowptr
     parts_s, parts_t, parts_w = [], [], []
-    pending = 0
-
-    def _fold():
-        nonlocal parts_s, parts_t, parts_w, pending
-        s, t, w = _aggregate(torch.cat(parts_s), torch.cat(parts_t),
-                             torch.cat(parts_w), gnc)
-        parts_s, parts_t, parts_w = [s], [t], [w]
-        pending = s.numel()
 
     for c0 in range(0, max(ne, 1), CH):
         c1 = min(c0 + CH, ne)
@@ -137,17 +125,8 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor,
         parts_s.append(cs)
         parts_t.append(ct)
         parts_w.append(cw)
-        pending += cs.numel()
-        if pending > MERGE_CAP:
-            _fold()
-    if parts_s:
-        if len(parts_s) > 1:
-            _fold()
-        s_agg, t_agg, w_agg = parts_s[0], parts_t[0], parts_w[0]
-    else:
-        s_agg = torch.zeros(0, dtype=torch.int64, device=dev)
-        t_agg = torch.zeros(0, dtype=torch.int64, device=dev)
-        w_agg = torch.zeros(0, dtype=dg.g.weights.dtype, device=dev)
+    s_agg, t_agg, w_agg = _fold_parts(parts_s, parts_t, parts_w, gnc, dev,
+                                      dg.g.weights.dtype)
     del parts_s, parts_t, parts_w
 
     if world > 1:
@@ -160,19 +139,88 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor,
                                   recv_counts=cnts)
         got_w = comm.all_to_all_v([w_agg[offs[p]:offs[p + 1]] for p in range(world)],
                                   recv_counts=cnts)
-        s_agg = torch.cat(got_s)
-        t_agg = torch.cat(got_t)
-        w_agg = torch.cat(got_w)
-        s_agg, t_agg, w_agg = _aggregate(s_agg, t_agg, w_agg, gnc)
+        # each peer's contribution arrives key-sorted: range-fold them
+        # (one flat sort would break the INT_MAX cap at s30-per-rank sizes)
+        s_agg, t_agg, w_agg = _fold_parts(
+            [g for g in got_s if g.numel()],
+            [g for g, gs in zip(got_t, got_s) if gs.numel()],
+            [g for g, gs in zip(got_w, got_s) if gs.numel()],
+            gnc, dev, dg.g.weights.dtype)
 
     nbase = new_part.base(rank)
     nv_new = new_part.nv_local(rank)
-    rowptr = torch.zeros(nv_new + 1, dtype=torch.int64, device=dev)
     if s_agg.numel():
-        rowptr[1:] = torch.cumsum(
-            torch.bincount(s_agg - nbase, minlength=nv_new), dim=0)
+        # s_agg is key-sorted, so the CSR rowptr is a direct searchsorted
+        # (bincount is capped at INT_MAX input elements; this is not)
+        rowptr = torch.searchsorted(
+            s_agg, torch.arange(nbase, nbase + nv_new + 1, device=dev))
+    else:
+        rowptr = torch.zeros(nv_new + 1, dtype=torch.int64, device=dev)
     new_g = Graph(rowptr, t_agg, w_agg)
     return DistGraph(new_g, new_part, rank), renum
+
+
+# one merge sort must stay under both the torch sort cap (INT_MAX elements)
+# and a sane transient-memory footprint (the cat + rocPRIM double buffers of
+# a 1.6G-entry fold were ~80 GB and OOM'd the s27 converged run)
+_FOLD_CAP = 1 << 29
+
+
+def _fold_parts(parts_s, parts_t, parts_w, gnc: int, dev, W):
+    """Merge key-sorted chunk aggregates into one globally aggregated,
+    key-sorted coarse edge list. Small totals: one cat + aggregate. Large
+    totals (s27/s30-per-rank): partition the KEY SPACE into ranges from the
+    largest part's quantiles and aggregate each range independently — no
+    single sort exceeds _FOLD_CAP elements and the total result may exceed
+    2^31 entries (only ever touched by elementwise/searchsorted ops after
+    this). A key appears at most once per part, so ranges can always be
+    split below the cap."""
+    if not parts_s:
+        return (torch.zeros(0, dtype=torch.int64, device=dev),
+                torch.zeros(0, dtype=torch.int64, device=dev),
+                torch.zeros(0, dtype=W, device=dev))
+    if len(parts_s) == 1:
+        return parts_s[0], parts_t[0], parts_w[0]
+    total = sum(int(p.numel()) for p in parts_s)
+    if total <= _FOLD_CAP:
+        return _aggregate(torch.cat(parts_s), torch.cat(parts_t),
+                          torch.cat(parts_w), gnc)
+    keys = [ps * gnc + pt for ps, pt in zip(parts_s, parts_t)]
+    big = max(range(len(keys)), key=lambda i: keys[i].numel())
+    n_ranges = (total + _FOLD_CAP // 2 - 1) // (_FOLD_CAP // 2)
+    bk = keys[big]
+    qpos = torch.linspace(0, bk.numel() - 1, n_ranges + 1,
+                          device=dev).to(torch.int64)[1:-1]
+    bounds = [None] + [int(bk[q]) for q in qpos] + [None]
+    out_s, out_t, out_w = [], [], []
+    for r in range(len(bounds) - 1):
+        lo, hi = bounds[r], bounds[r + 1]
+        sl_s, sl_t, sl_w = [], [], []
+        for ps, pt, pw, pk in zip(parts_s, parts_t, parts_w, keys):
+            a = 0 if lo is None else int(torch.searchsorted(
+                pk, torch.tensor(lo, device=dev)))
+            b = pk.numel() if hi is None else int(torch.searchsorted(
+                pk, torch.tensor(hi, device=dev)))
+            if b > a:
+                sl_s.append(ps[a:b])
+                sl_t.append(pt[a:b])
+                sl_w.append(pw[a:b])
+        if not sl_s:
+            continue
+        n_r = sum(int(x.numel()) for x in sl_s)
+        if len(sl_s) == 1:
+            # single sorted, already-deduped slice: pass through
+            cs, ct, cw = sl_s[0], sl_t[0], sl_w[0]
+        elif n_r > _FOLD_CAP:
+            # rare skewed range: recurse with the slice list
+            cs, ct, cw = _fold_parts(sl_s, sl_t, sl_w, gnc, dev, W)
+        else:
+            cs, ct, cw = _aggregate(torch.cat(sl_s), torch.cat(sl_t),
+                                    torch.cat(sl_w), gnc)
+        out_s.append(cs)
+        out_t.append(ct)
+        out_w.append(cw)
+    return torch.cat(out_s), torch.cat(out_t), torch.cat(out_w)
 
 
 def _aggregate(s: torch.Tensor, t: torch.Tensor, w: torch.Tensor, gnc: int):
