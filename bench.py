@@ -139,9 +139,10 @@ def main():
         # release_tails). The halo travels through a single-ref handoff so
         # louvain() can drop it once phase 0's coarsening is done.
         conv_halo = [state.halo]
-        del state
+        del state, step  # step's closure cell would otherwise keep state
         if device.type == "cuda" and comm.world == 1:
             dg.g.release_tails()
+            torch.cuda.empty_cache()
         t0 = time.perf_counter()
         try:
             res = louvain(dg, comm, LouvainConfig(backend=args.backend),
@@ -164,6 +165,25 @@ def main():
         except Exception as e:  # the PRIMARY timed metric must still print
             conv = {"error": f"{type(e).__name__}: {e}"[:300]}
             _p(f"converged run failed: {conv['error']}")
+            import traceback
+            traceback.print_exc(file=sys.stderr)
+            if device.type == "cuda":
+                # live-tensor census for memory debugging
+                import gc
+                sizes = {}
+                for o in gc.get_objects():
+                    try:
+                        if torch.is_tensor(o) and o.is_cuda:
+                            k = (tuple(o.shape), str(o.dtype))
+                            sizes[k] = sizes.get(k, 0) + \
+                                o.numel() * o.element_size()
+                    except Exception:
+                        pass
+                top = sorted(sizes.items(), key=lambda kv: -kv[1])[:15]
+                for (shape, dt), b in top:
+                    _p(f"live {b/2**30:7.2f} GiB  {dt}  {shape}")
+                _p(f"allocated {torch.cuda.memory_allocated()/2**30:.1f} "
+                   f"GiB reserved {torch.cuda.memory_reserved()/2**30:.1f}")
 
     if comm.rank == 0:
         out = {
