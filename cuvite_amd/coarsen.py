@@ -106,10 +106,14 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor,
     # chunked over the edge list: torch sort/advanced-indexing are capped at
     # INT_MAX elements and the per-chunk aggregates are far smaller than the
     # fine edge list, so partial aggregation also bounds peak memory
+    # coarse edges live as packed keys s*gnc+t (16 B/entry with the weight
+    # instead of 24 — the s component is never materialized: routing, the
+    # CSR rowptr and the tails all derive from the key directly)
     ne = dg.g.ne
     CH = 1 << 28
     rowptr = dg.g.rowptr
-    parts_s, parts_t, parts_w = [], [], []
+    W = dg.g.weights.dtype
+    parts_k, parts_w = [], []
 
     for c0 in range(0, max(ne, 1), CH):
         c1 = min(c0 + CH, ne)
@@ -118,44 +122,42 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor,
         eidx = torch.arange(c0, c1, device=dev)
         seg = torch.searchsorted(rowptr, eidx, right=True) - 1
         del eidx
-        s_new = new_of_all[seg]
+        key = new_of_all[seg] * gnc
         del seg
-        t_new = new_of_all[halo.tails_dense[c0:c1].to(torch.int64)]
-        cs, ct, cw = _aggregate(s_new, t_new, dg.g.weights[c0:c1], gnc)
-        parts_s.append(cs)
-        parts_t.append(ct)
+        key += new_of_all[halo.tails_dense[c0:c1].to(torch.int64)]
+        ck, cw = _aggregate_keys(key, dg.g.weights[c0:c1])
+        del key
+        parts_k.append(ck)
         parts_w.append(cw)
-    s_agg, t_agg, w_agg = _fold_parts(parts_s, parts_t, parts_w, gnc, dev,
-                                      dg.g.weights.dtype)
-    del parts_s, parts_t, parts_w
+    k_agg, w_agg = _fold_keys(parts_k, parts_w, dev, W)
+    del parts_k, parts_w
 
     if world > 1:
         npdev = new_part.parts.to(dev)
-        offs = torch.searchsorted(s_agg, npdev)
-        sp = [s_agg[offs[p]:offs[p + 1]] for p in range(world)]
-        got_s = comm.all_to_all_v(sp)
-        cnts = [int(g.numel()) for g in got_s]
-        got_t = comm.all_to_all_v([t_agg[offs[p]:offs[p + 1]] for p in range(world)],
-                                  recv_counts=cnts)
-        got_w = comm.all_to_all_v([w_agg[offs[p]:offs[p + 1]] for p in range(world)],
-                                  recv_counts=cnts)
+        offs = torch.searchsorted(k_agg, npdev * gnc)
+        got_k = comm.all_to_all_v(
+            [k_agg[offs[p]:offs[p + 1]] for p in range(world)])
+        cnts = [int(g.numel()) for g in got_k]
+        got_w = comm.all_to_all_v(
+            [w_agg[offs[p]:offs[p + 1]] for p in range(world)],
+            recv_counts=cnts)
         # each peer's contribution arrives key-sorted: range-fold them
         # (one flat sort would break the INT_MAX cap at s30-per-rank sizes)
-        s_agg, t_agg, w_agg = _fold_parts(
-            [g for g in got_s if g.numel()],
-            [g for g, gs in zip(got_t, got_s) if gs.numel()],
-            [g for g, gs in zip(got_w, got_s) if gs.numel()],
-            gnc, dev, dg.g.weights.dtype)
+        k_agg, w_agg = _fold_keys(got_k, got_w, dev, W)
 
     nbase = new_part.base(rank)
     nv_new = new_part.nv_local(rank)
-    if s_agg.numel():
-        # s_agg is key-sorted, so the CSR rowptr is a direct searchsorted
+    if k_agg.numel():
+        # key-sorted, so the CSR rowptr is a direct searchsorted
         # (bincount is capped at INT_MAX input elements; this is not)
         rowptr = torch.searchsorted(
-            s_agg, torch.arange(nbase, nbase + nv_new + 1, device=dev))
+            k_agg, torch.arange(nbase, nbase + nv_new + 1,
+                                device=dev) * gnc)
+        t_agg = k_agg % gnc
     else:
         rowptr = torch.zeros(nv_new + 1, dtype=torch.int64, device=dev)
+        t_agg = torch.zeros(0, dtype=torch.int64, device=dev)
+    del k_agg
     new_g = Graph(rowptr, t_agg, w_agg)
     return DistGraph(new_g, new_part, rank), renum
 
@@ -166,82 +168,77 @@ def coarsen(dg: DistGraph, comm: Comm, cvect: torch.Tensor,
 _FOLD_CAP = 1 << 29
 
 
-def _fold_parts(parts_s, parts_t, parts_w, gnc: int, dev, W):
-    """Merge key-sorted chunk aggregates into one globally aggregated,
-    key-sorted coarse edge list. Small totals: one cat + aggregate. Large
+def _fold_keys(parts_k, parts_w, dev, W):
+    """Merge key-sorted (key, weight) chunk aggregates into one globally
+    aggregated, key-sorted pair. Small totals: one cat + aggregate. Large
     totals (s27/s30-per-rank): partition the KEY SPACE into ranges from the
     largest part's quantiles and aggregate each range independently — no
     single sort exceeds _FOLD_CAP elements and the total result may exceed
     2^31 entries (only ever touched by elementwise/searchsorted ops after
     this). A key appears at most once per part, so ranges can always be
     split below the cap."""
-    if not parts_s:
+    parts_k = [p for p in parts_k if p.numel()]
+    parts_w = [p for p in parts_w if p.numel()]
+    if not parts_k:
         return (torch.zeros(0, dtype=torch.int64, device=dev),
-                torch.zeros(0, dtype=torch.int64, device=dev),
                 torch.zeros(0, dtype=W, device=dev))
-    if len(parts_s) == 1:
-        return parts_s[0], parts_t[0], parts_w[0]
-    total = sum(int(p.numel()) for p in parts_s)
+    if len(parts_k) == 1:
+        return parts_k[0], parts_w[0]
+    total = sum(int(p.numel()) for p in parts_k)
     if total <= _FOLD_CAP:
-        return _aggregate(torch.cat(parts_s), torch.cat(parts_t),
-                          torch.cat(parts_w), gnc)
-    keys = [ps * gnc + pt for ps, pt in zip(parts_s, parts_t)]
-    big = max(range(len(keys)), key=lambda i: keys[i].numel())
+        return _aggregate_keys(torch.cat(parts_k), torch.cat(parts_w))
+    big = max(range(len(parts_k)), key=lambda i: parts_k[i].numel())
     n_ranges = (total + _FOLD_CAP // 2 - 1) // (_FOLD_CAP // 2)
-    bk = keys[big]
+    bk = parts_k[big]
     qpos = torch.linspace(0, bk.numel() - 1, n_ranges + 1,
                           device=dev).to(torch.int64)[1:-1]
     bounds = [None] + [int(bk[q]) for q in qpos] + [None]
-    out_s, out_t, out_w = [], [], []
+    out_k, out_w = [], []
     for r in range(len(bounds) - 1):
         lo, hi = bounds[r], bounds[r + 1]
-        sl_s, sl_t, sl_w = [], [], []
-        for ps, pt, pw, pk in zip(parts_s, parts_t, parts_w, keys):
+        sl_k, sl_w = [], []
+        for pk, pw in zip(parts_k, parts_w):
             a = 0 if lo is None else int(torch.searchsorted(
                 pk, torch.tensor(lo, device=dev)))
             b = pk.numel() if hi is None else int(torch.searchsorted(
                 pk, torch.tensor(hi, device=dev)))
             if b > a:
-                sl_s.append(ps[a:b])
-                sl_t.append(pt[a:b])
+                sl_k.append(pk[a:b])
                 sl_w.append(pw[a:b])
-        if not sl_s:
+        if not sl_k:
             continue
-        n_r = sum(int(x.numel()) for x in sl_s)
-        if len(sl_s) == 1:
+        n_r = sum(int(x.numel()) for x in sl_k)
+        if len(sl_k) == 1:
             # single sorted, already-deduped slice: pass through
-            cs, ct, cw = sl_s[0], sl_t[0], sl_w[0]
+            ck, cw = sl_k[0], sl_w[0]
         elif n_r > _FOLD_CAP:
             # rare skewed range: recurse with the slice list
-            cs, ct, cw = _fold_parts(sl_s, sl_t, sl_w, gnc, dev, W)
+            ck, cw = _fold_keys(sl_k, sl_w, dev, W)
         else:
-            cs, ct, cw = _aggregate(torch.cat(sl_s), torch.cat(sl_t),
-                                    torch.cat(sl_w), gnc)
-        out_s.append(cs)
-        out_t.append(ct)
+            ck, cw = _aggregate_keys(torch.cat(sl_k), torch.cat(sl_w))
+        out_k.append(ck)
         out_w.append(cw)
-    return torch.cat(out_s), torch.cat(out_t), torch.cat(out_w)
+    return torch.cat(out_k), torch.cat(out_w)
 
 
-def _aggregate(s: torch.Tensor, t: torch.Tensor, w: torch.Tensor, gnc: int):
-    """Merge duplicate (s, t) directed edges, summing weights; returns sorted
-    by (s, t). Keys fit int64 because gnc < 2^31."""
-    key = s * gnc + t
+def _aggregate_keys(key: torch.Tensor, w: torch.Tensor):
+    """Sum weights of duplicate keys; returns (uniq keys sorted, sums),
+    both exactly m-sized (full-buffer views are CLONED — a [:m] slice of the
+    rocPRIM output would pin the n-sized buffer, which leaked 64 GB across
+    the s27 chunk aggregates before this was a clone)."""
     if key.is_cuda:
         from . import ops
         if ops.available():
-            # narrow-bit rocPRIM radix sort (only bits of gnc^2, ~50 at s26
-            # vs 64 for a generic int64 sort) + reduce_by_key — replaces the
-            # sort / gather / unique_consecutive / double-cumsum chain that
-            # dominated the 5.9 s phase-0 rebuild (profiles/NEXT.md)
-            end_bit = max(1, (gnc * gnc - 1).bit_length())
+            # narrow-bit rocPRIM radix sort (only the bits the key uses,
+            # ~50 at s26 vs 64 for a generic int64 sort) + reduce_by_key —
+            # replaces the sort / gather / unique_consecutive /
+            # double-cumsum chain that dominated the 5.9 s phase-0 rebuild
+            end_bit = int(key.max()).bit_length() if key.numel() else 1
             w64 = w if w.dtype == torch.float64 else w.to(torch.float64)
             uniq, sums, cnt = ops._require().sort_reduce_pairs(
-                key, w64, min(end_bit, 64))
+                key, w64, max(1, min(end_bit, 64)))
             m = int(cnt[0])
-            uniq = uniq[:m]
-            sums = sums[:m]
-            return uniq // gnc, uniq % gnc, sums.to(w.dtype)
+            return uniq[:m].clone(), sums[:m].to(w.dtype).clone()
     key_s, order = torch.sort(key)
     w_s = w[order]
     uniq, counts = torch.unique_consecutive(key_s, return_counts=True)
@@ -251,7 +248,14 @@ def _aggregate(s: torch.Tensor, t: torch.Tensor, w: torch.Tensor, gnc: int):
     cs = torch.cumsum(w_s.to(torch.float64), dim=0)
     w_out = cs[ends].clone()
     w_out[1:] -= cs[ends[:-1]]
-    return uniq // gnc, uniq % gnc, w_out.to(w.dtype)
+    return uniq, w_out.to(w.dtype)
+
+
+def _aggregate(s: torch.Tensor, t: torch.Tensor, w: torch.Tensor, gnc: int):
+    """Merge duplicate (s, t) directed edges, summing weights; returns sorted
+    by (s, t). Keys fit int64 because gnc < 2^31."""
+    uniq, sums = _aggregate_keys(s * gnc + t, w)
+    return uniq // gnc, uniq % gnc, sums
 
 
 def remap_labels(dg: DistGraph, comm: Comm, assign: torch.Tensor,

@@ -139,7 +139,7 @@ def test_timers_accumulate():
     assert "x=" in ts.summary()
 
 
-def test_fold_parts_range_partition_matches_flat():
+def test_fold_keys_range_partition_matches_flat():
     """Key-range folding (the >INT_MAX-safe coarse-edge merge) must equal a
     flat cat+aggregate, including with a cap small enough to force many
     ranges and recursion."""
@@ -152,18 +152,15 @@ def test_fold_parts_range_partition_matches_flat():
         s = torch.randint(0, gnc, (n,), dtype=torch.int64)
         t = torch.randint(0, gnc, (n,), dtype=torch.int64)
         w = torch.rand(n, dtype=torch.float64)
-        chunks.append(C._aggregate(s, t, w, gnc))
-    ref = C._aggregate(torch.cat([c[0] for c in chunks]),
-                       torch.cat([c[1] for c in chunks]),
-                       torch.cat([c[2] for c in chunks]), gnc)
+        chunks.append(C._aggregate_keys(s * gnc + t, w))
+    ref = C._aggregate_keys(torch.cat([c[0] for c in chunks]),
+                            torch.cat([c[1] for c in chunks]))
     old = C._FOLD_CAP
     try:
         C._FOLD_CAP = 1024  # force range partitioning + recursion
-        got = C._fold_parts([c[0] for c in chunks], [c[1] for c in chunks],
-                            [c[2] for c in chunks], gnc,
-                            torch.device("cpu"), torch.float64)
+        got = C._fold_keys([c[0] for c in chunks], [c[1] for c in chunks],
+                           torch.device("cpu"), torch.float64)
     finally:
         C._FOLD_CAP = old
     assert torch.equal(got[0], ref[0])
-    assert torch.equal(got[1], ref[1])
-    assert torch.allclose(got[2], ref[2])
+    assert torch.allclose(got[1], ref[1])
