@@ -44,12 +44,15 @@ def _hub_cut() -> int:
     hub pipeline. The 8192-slot table stays collision-safe up to 8000
     distinct neighbors (load <= 0.98 worst case, ~0.8 typical); R-MAT's
     lumpy degree spectrum clusters many "hubs" just above 4096, so raising
-    the cut keeps them on the (cheaper) LDS path. A/B-tunable without a
-    rebuild: CUVITE_HUB_CUT, default 4096 (the reference's CUT_SIZE)."""
+    the cut keeps them on the (cheaper) LDS path. A/B at s26 (gpurun r2c):
+    cut 4096 = 150.1 ms/step, 6144 = 136.0, 8000 = 136.7 (identical
+    trajectories; hub count 84k -> 18k, hub sort 74 -> 43 ms). Default 6144;
+    CUVITE_HUB_CUT overrides (the reference's CUT_SIZE was 4096,
+    louvain_cuda_constants.cuh:34)."""
     try:
-        cut = int(os.environ.get("CUVITE_HUB_CUT", "4096"))
+        cut = int(os.environ.get("CUVITE_HUB_CUT", "6144"))
     except ValueError:
-        cut = 4096
+        cut = 6144
     return max(2049, min(cut, 8000))
 
 
