@@ -365,3 +365,38 @@ def test_release_tails_louvain_still_runs():
     got = louvain(dg, Comm(dev), LouvainConfig(backend="hip"), halo=h)
     assert got.modularity == ref.modularity
     assert torch.equal(got.communities, ref.communities)
+
+
+def test_hub_cut_parity():
+    """CUVITE_HUB_CUT moves vertices between the LDS block class and the
+    segsort hub pipeline; results must be identical for any cut."""
+    from cuvite_amd import ops
+    torch.manual_seed(13)
+    nv = 4096
+    # several vertices with degrees straddling the 4096..8000 band
+    src, dst = [], []
+    for hub, d in ((0, 7000), (1, 5000), (2, 4500), (3, 9000)):
+        src += [hub] * d
+        dst += torch.randint(4, nv, (d,)).tolist()
+    src += torch.randint(4, nv, (nv * 4,)).tolist()
+    dst += torch.randint(4, nv, (nv * 4,)).tolist()
+    s = torch.tensor(src, dtype=torch.int64)
+    d = torch.tensor(dst, dtype=torch.int64)
+    w = torch.ones(s.numel(), dtype=torch.float64)
+    g = Graph.from_edge_tuples(nv, s, d, w)
+    inp = _inputs(g, torch.device("cuda:0"), "random", seed=6)
+    outs = []
+    for cut in ("4096", "6144", "8000"):
+        os.environ["CUVITE_HUB_CUT"] = cut
+        try:
+            ops._bucket_cache.clear()
+            ops._hub_static_cache.clear()
+            ops._hub_groups_cache.clear()
+            outs.append(ops.local_move(inp))
+        finally:
+            del os.environ["CUVITE_HUB_CUT"]
+    ops._bucket_cache.clear()
+    ops._hub_static_cache.clear()
+    for t2, cw2 in outs[1:]:
+        assert torch.equal(outs[0][0], t2)
+        assert torch.equal(outs[0][1], cw2)
