@@ -58,12 +58,14 @@ def distance1_coloring(dg: DistGraph, comm: Comm, n_hash: int = 4,
     base = dg.base
     if halo is None:
         halo = build_halo(dg, comm)
-    tails = halo.tails_dense.to(torch.int64)
-    seg = torch.repeat_interleave(torch.arange(nv, device=dev), dg.g.degrees())
+    rowptr = dg.g.rowptr
+    ne = dg.g.ne
     gid_all = torch.cat([torch.arange(base, dg.bound, device=dev), halo.ghosts])
-    tail_gid = gid_all[tails]
-    not_self = tail_gid != (seg + base)
     tnv = dg.nv_global
+    # edges are processed in chunks: boolean compaction over a >INT_MAX edge
+    # array overflows torch's internal indexing (hit at R-MAT s26), and
+    # chunking also avoids materializing 17 GB seg/tail_gid arrays
+    CH = 1 << 28
 
     colors = torch.full((nv,), -1, dtype=torch.int64, device=dev)
     seed = 1012
@@ -75,24 +77,38 @@ def distance1_coloring(dg: DistGraph, comm: Comm, n_hash: int = 4,
         # competing = uncolored at round start (local + ghosts)
         ghost_colors = exchange_ghost_labels(halo, colors)
         uncolored_all = torch.cat([colors, ghost_colors]) < 0
-        cand_edges = not_self & uncolored_all[tails] & (colors[seg] < 0)
-        e_seg = seg[cand_edges]
-        e_tail_gid = tail_gid[cand_edges]
+        # strict min / strict max of each hash over competing neighbors,
+        # accumulated chunkwise over the edge list
+        mns = [torch.full((nv,), 1 << 33, dtype=torch.int64, device=dev)
+               for _ in range(n_hash)]
+        mxs = [torch.full((nv,), -1, dtype=torch.int64, device=dev)
+               for _ in range(n_hash)]
+        for c0 in range(0, ne, CH):
+            c1 = min(c0 + CH, ne)
+            eidx = torch.arange(c0, c1, device=dev)
+            seg_c = torch.searchsorted(rowptr, eidx, right=True) - 1
+            del eidx
+            tails_c = halo.tails_dense[c0:c1].to(torch.int64)
+            tail_gid_c = gid_all[tails_c]
+            cand = (tail_gid_c != (seg_c + base)) \
+                & uncolored_all[tails_c] & (colors[seg_c] < 0)
+            del tails_c
+            e_seg = seg_c[cand]
+            e_tail_gid = tail_gid_c[cand]
+            del seg_c, tail_gid_c, cand
+            if not e_seg.numel():
+                continue
+            for t in range(n_hash):
+                jh = _hash(e_tail_gid, seed + 1043 * t)
+                mns[t].scatter_reduce_(0, e_seg, jh, reduce="amin")
+                mxs[t].scatter_reduce_(0, e_seg, jh, reduce="amax")
 
         avail = torch.zeros(nv, 2 * n_hash, dtype=torch.bool, device=dev)
         vgid = torch.arange(base, dg.bound, device=dev)
         for t in range(n_hash):
-            hseed = seed + 1043 * t
-            vh = _hash(vgid, hseed)
-            jh = _hash(e_tail_gid, hseed)
-            # strict min / strict max over competing neighbors
-            mn = torch.full((nv,), 1 << 33, dtype=torch.int64, device=dev)
-            mx = torch.full((nv,), -1, dtype=torch.int64, device=dev)
-            if e_seg.numel():
-                mn.scatter_reduce_(0, e_seg, jh, reduce="amin")
-                mx.scatter_reduce_(0, e_seg, jh, reduce="amax")
-            avail[:, 2 * t] = vh < mn
-            avail[:, 2 * t + 1] = vh > mx
+            vh = _hash(vgid, seed + 1043 * t)
+            avail[:, 2 * t] = vh < mns[t]
+            avail[:, 2 * t + 1] = vh > mxs[t]
 
         uncolored = colors < 0
         navail = avail.sum(dim=1)
@@ -129,14 +145,22 @@ def check_coloring(dg: DistGraph, comm: Comm, colors: torch.Tensor,
     nv = dg.nv
     if halo is None:
         halo = build_halo(dg, comm)
-    tails = halo.tails_dense.to(torch.int64)
-    seg = torch.repeat_interleave(torch.arange(nv, device=dev), dg.g.degrees())
     ghost_colors = exchange_ghost_labels(halo, colors)
     call = torch.cat([colors, ghost_colors])
     gid_all = torch.cat([torch.arange(dg.base, dg.bound, device=dev), halo.ghosts])
-    not_self = gid_all[tails] != (seg + dg.base)
-    same = not_self & (call[tails] == colors[seg])
-    if exclude_color >= 0:
-        same &= colors[seg] != exclude_color
-    conflicts = int(same.sum())
+    rowptr = dg.g.rowptr
+    ne = dg.g.ne
+    CH = 1 << 28  # chunked like distance1_coloring (>INT_MAX edge lists)
+    conflicts = 0
+    for c0 in range(0, ne, CH):
+        c1 = min(c0 + CH, ne)
+        eidx = torch.arange(c0, c1, device=dev)
+        seg_c = torch.searchsorted(rowptr, eidx, right=True) - 1
+        del eidx
+        tails_c = halo.tails_dense[c0:c1].to(torch.int64)
+        same = (gid_all[tails_c] != (seg_c + dg.base)) \
+            & (call[tails_c] == colors[seg_c])
+        if exclude_color >= 0:
+            same &= colors[seg_c] != exclude_color
+        conflicts += int(same.sum())
     return int(comm.allreduce_scalar(float(conflicts)))
