@@ -73,35 +73,53 @@ def distance1_coloring(dg: DistGraph, comm: Comm, n_hash: int = 4,
     last_count = 0
     target = (tnv * MAX_COVG) // 100
 
+    use_hip = dev.type == "cuda"
+    if use_hip:
+        from . import ops
+        use_hip = ops.available()
+
     while True:
         # competing = uncolored at round start (local + ghosts)
         ghost_colors = exchange_ghost_labels(halo, colors)
         uncolored_all = torch.cat([colors, ghost_colors]) < 0
-        # strict min / strict max of each hash over competing neighbors,
-        # accumulated chunkwise over the edge list
-        mns = [torch.full((nv,), 1 << 33, dtype=torch.int64, device=dev)
-               for _ in range(n_hash)]
-        mxs = [torch.full((nv,), -1, dtype=torch.int64, device=dev)
-               for _ in range(n_hash)]
-        for c0 in range(0, ne, CH):
-            c1 = min(c0 + CH, ne)
-            eidx = torch.arange(c0, c1, device=dev)
-            seg_c = torch.searchsorted(rowptr, eidx, right=True) - 1
-            del eidx
-            tails_c = halo.tails_dense[c0:c1].to(torch.int64)
-            tail_gid_c = gid_all[tails_c]
-            cand = (tail_gid_c != (seg_c + base)) \
-                & uncolored_all[tails_c] & (colors[seg_c] < 0)
-            del tails_c
-            e_seg = seg_c[cand]
-            e_tail_gid = tail_gid_c[cand]
-            del seg_c, tail_gid_c, cand
-            if not e_seg.numel():
-                continue
-            for t in range(n_hash):
-                jh = _hash(e_tail_gid, seed + 1043 * t)
-                mns[t].scatter_reduce_(0, e_seg, jh, reduce="amin")
-                mxs[t].scatter_reduce_(0, e_seg, jh, reduce="amax")
+        # strict min / strict max of each hash over competing neighbors
+        if use_hip:
+            # one wave per vertex, all hashes in a single edge pass
+            from . import ops
+            seeds = torch.tensor(
+                [(seed + 1043 * t) & 0xFFFFFFFF for t in range(n_hash)],
+                dtype=torch.int64, device=dev)
+            mn2, mx2 = ops._require().coloring_minmax(
+                rowptr, halo.tails_dense, gid_all, uncolored_all, colors,
+                base, seeds)
+            mns = [mn2[:, t] for t in range(n_hash)]
+            mxs = [mx2[:, t] for t in range(n_hash)]
+        else:
+            # chunked torch path (boolean compaction over >INT_MAX edges
+            # overflows torch's internal indexing, hence CH slices)
+            mns = [torch.full((nv,), 1 << 33, dtype=torch.int64, device=dev)
+                   for _ in range(n_hash)]
+            mxs = [torch.full((nv,), -1, dtype=torch.int64, device=dev)
+                   for _ in range(n_hash)]
+            for c0 in range(0, ne, CH):
+                c1 = min(c0 + CH, ne)
+                eidx = torch.arange(c0, c1, device=dev)
+                seg_c = torch.searchsorted(rowptr, eidx, right=True) - 1
+                del eidx
+                tails_c = halo.tails_dense[c0:c1].to(torch.int64)
+                tail_gid_c = gid_all[tails_c]
+                cand = (tail_gid_c != (seg_c + base)) \
+                    & uncolored_all[tails_c] & (colors[seg_c] < 0)
+                del tails_c
+                e_seg = seg_c[cand]
+                e_tail_gid = tail_gid_c[cand]
+                del seg_c, tail_gid_c, cand
+                if not e_seg.numel():
+                    continue
+                for t in range(n_hash):
+                    jh = _hash(e_tail_gid, seed + 1043 * t)
+                    mns[t].scatter_reduce_(0, e_seg, jh, reduce="amin")
+                    mxs[t].scatter_reduce_(0, e_seg, jh, reduce="amax")
 
         avail = torch.zeros(nv, 2 * n_hash, dtype=torch.bool, device=dev)
         vgid = torch.arange(base, dg.bound, device=dev)

@@ -63,6 +63,10 @@ int hub_argmax_splits();
 template <typename W>
 void launch_apply_deltas(const int64_t*, const int64_t*, const W*, int64_t,
                          int64_t, int64_t, int64_t*, W*, hipStream_t);
+void launch_coloring_minmax(const int64_t*, const int32_t*, const int64_t*,
+                            const bool*, const int64_t*, int64_t, int64_t,
+                            const int64_t*, int, int64_t*, int64_t*,
+                            hipStream_t);
 
 }  // namespace cuvite
 
@@ -363,6 +367,36 @@ void apply_deltas_(at::Tensor target, at::Tensor curr, at::Tensor v_degree,
   C10_HIP_CHECK(hipGetLastError());
 }
 
+// One coloring round's strict min/max of each hash over competing
+// neighbors, per vertex (ref distColoringIteration, coloring.cpp:87-202).
+std::vector<at::Tensor> coloring_minmax(at::Tensor rowptr, at::Tensor tails,
+                                        at::Tensor gid_all,
+                                        at::Tensor uncolored,
+                                        at::Tensor colors, int64_t base,
+                                        at::Tensor seeds) {
+  CHECK_DEV(rowptr); CHECK_CONT(rowptr);
+  CHECK_DEV(tails); CHECK_CONT(tails);
+  CHECK_DEV(gid_all); CHECK_CONT(gid_all);
+  CHECK_DEV(uncolored); CHECK_CONT(uncolored);
+  CHECK_DEV(colors); CHECK_CONT(colors);
+  TORCH_CHECK(tails.scalar_type() == at::kInt);
+  TORCH_CHECK(uncolored.scalar_type() == at::kBool);
+  TORCH_CHECK(seeds.scalar_type() == at::kLong && seeds.is_cuda());
+  const int n_hash = (int)seeds.numel();
+  TORCH_CHECK(1 <= n_hash && n_hash <= 8, "n_hash must be in [1,8]");
+  const int64_t nv = rowptr.numel() - 1;
+  auto mn = at::empty({nv, n_hash}, rowptr.options());
+  auto mx = at::empty({nv, n_hash}, rowptr.options());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  cuvite::launch_coloring_minmax(
+      rowptr.data_ptr<int64_t>(), tails.data_ptr<int32_t>(),
+      gid_all.data_ptr<int64_t>(), uncolored.data_ptr<bool>(),
+      colors.data_ptr<int64_t>(), nv, base, seeds.data_ptr<int64_t>(),
+      n_hash, mn.data_ptr<int64_t>(), mx.data_ptr<int64_t>(), stream);
+  C10_HIP_CHECK(hipGetLastError());
+  return {mn, mx};
+}
+
 at::Tensor row_sum(at::Tensor rowptr, at::Tensor weights) {
   CHECK_DEV(rowptr); CHECK_CONT(rowptr);
   CHECK_DEV(weights); CHECK_CONT(weights);
@@ -395,6 +429,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sort_reduce_pairs", &sort_reduce_pairs,
         "narrow-bit radix sort + reduce_by_key coarse-edge aggregate "
         "(rocPRIM)");
+  m.def("coloring_minmax", &coloring_minmax,
+        "per-vertex multi-hash min/max over competing neighbors (HIP)");
   m.def("hub_moves", &hub_moves,
         "full device-side hub move: segsort + reduce_by_key + argmax");
 }

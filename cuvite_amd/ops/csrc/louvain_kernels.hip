@@ -445,6 +445,99 @@ __global__ void csr_place_kernel(const int64_t* __restrict__ src,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Distance-1 coloring round: per-vertex strict min/max of every hash over
+// COMPETING neighbors (uncolored at round start, not self), one wave per
+// vertex with all n_hash hashes in a single edge pass and a wave reduction —
+// no atomics, no edge-list materialization. Replaces 2*n_hash torch scatter
+// passes over the full edge list per round (13.4 s for the whole coloring
+// at R-MAT s26). Hash = the reference's 32-bit mix (coloring.cpp:74-85).
+// ---------------------------------------------------------------------------
+
+DEV_INLINE uint32_t color_hash(uint32_t a, uint32_t seed) {
+  a ^= seed;
+  a = (a + 0x7ED55D16u) + (a << 12);
+  a = (a ^ 0xC761C23Cu) + (a >> 19);
+  a = (a + 0x165667B1u) + (a << 5);
+  a = (a ^ 0xD3A2646Cu) + (a << 9);
+  a = (a + 0xFD7046C5u) + (a << 3);
+  a = (a ^ 0xB55A4F09u) + (a >> 16);
+  return a;
+}
+
+constexpr int MAX_NHASH = 8;
+
+template <int BLOCK>
+__global__ __launch_bounds__(BLOCK) void coloring_minmax_kernel(
+    const int64_t* __restrict__ rowptr, const int32_t* __restrict__ tails,
+    const int64_t* __restrict__ gid_all, const bool* __restrict__ uncolored,
+    const int64_t* __restrict__ colors, int64_t nv, int64_t base,
+    const int64_t* __restrict__ seeds, int n_hash,
+    int64_t* __restrict__ mn_out, int64_t* __restrict__ mx_out) {
+  constexpr int WAVES = BLOCK / 64;
+  const int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+  const int64_t vstride = (int64_t)gridDim.x * WAVES;
+  for (int64_t v = (int64_t)blockIdx.x * WAVES + wave; v < nv; v += vstride) {
+    uint32_t mn[MAX_NHASH], mx[MAX_NHASH];
+#pragma unroll
+    for (int t = 0; t < MAX_NHASH; t++) {
+      mn[t] = 0xFFFFFFFFu;
+      mx[t] = 0u;
+    }
+    bool any = false;
+    if (colors[v] < 0) {
+      const int64_t e0 = rowptr[v], e1 = rowptr[v + 1];
+      const int64_t vg = v + base;
+      for (int64_t e = e0 + lane; e < e1; e += 64) {
+        const int32_t td = tails[e];
+        if (!uncolored[td]) continue;
+        const int64_t g = gid_all[td];
+        if (g == vg) continue;
+        any = true;
+        for (int t = 0; t < n_hash; t++) {
+          const uint32_t h = color_hash((uint32_t)g, (uint32_t)seeds[t]);
+          mn[t] = h < mn[t] ? h : mn[t];
+          mx[t] = h > mx[t] ? h : mx[t];
+        }
+      }
+    }
+    any = __any(any);
+    for (int t = 0; t < n_hash; t++) {
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) {
+        const uint32_t omn = (uint32_t)__shfl_down((int)mn[t], off, 64);
+        const uint32_t omx = (uint32_t)__shfl_down((int)mx[t], off, 64);
+        mn[t] = omn < mn[t] ? omn : mn[t];
+        mx[t] = omx > mx[t] ? omx : mx[t];
+      }
+    }
+    if (lane == 0) {
+      for (int t = 0; t < n_hash; t++) {
+        // no competitor: same sentinels as the torch path (1<<33 / -1)
+        mn_out[v * n_hash + t] =
+            any ? (int64_t)mn[t] : ((int64_t)1 << 33);
+        mx_out[v * n_hash + t] = any ? (int64_t)mx[t] : (int64_t)-1;
+      }
+    }
+  }
+}
+
+void launch_coloring_minmax(const int64_t* rowptr, const int32_t* tails,
+                            const int64_t* gid_all, const bool* uncolored,
+                            const int64_t* colors, int64_t nv, int64_t base,
+                            const int64_t* seeds, int n_hash, int64_t* mn_out,
+                            int64_t* mx_out, hipStream_t stream) {
+  if (nv == 0) return;
+  constexpr int BLOCK = 256;
+  constexpr int WAVES = BLOCK / 64;
+  int64_t grid = (nv + WAVES - 1) / WAVES;
+  if (grid > 1048576) grid = 1048576;
+  hipLaunchKernelGGL((coloring_minmax_kernel<BLOCK>), dim3((uint32_t)grid),
+                     dim3(BLOCK), 0, stream, rowptr, tails, gid_all,
+                     uncolored, colors, nv, base, seeds, n_hash, mn_out,
+                     mx_out);
+}
+
 // Per-vertex weighted degree: one wave per vertex, lane-strided row scan
 // (ref distSumVertexDegree, louvain.cpp:2126-2151).
 template <typename W, int BLOCK>

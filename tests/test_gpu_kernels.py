@@ -400,3 +400,18 @@ def test_hub_cut_parity():
     for t2, cw2 in outs[1:]:
         assert torch.equal(outs[0][0], t2)
         assert torch.equal(outs[0][1], cw2)
+
+
+def test_coloring_gpu_matches_cpu_exact():
+    """The fused coloring_minmax kernel path must produce EXACTLY the CPU
+    coloring (integer min/max, no fp, no atomics)."""
+    from cuvite_amd.coloring import distance1_coloring, check_coloring
+    from cuvite_amd.parallel import Comm
+    g = rmat_graph(11, 16, seed=6)
+    c_cpu, n_cpu = distance1_coloring(single_partition(g), Comm(), n_hash=4)
+    dev = torch.device("cuda:0")
+    dgg = single_partition(g.to(dev))
+    c_gpu, n_gpu = distance1_coloring(dgg, Comm(dev), n_hash=4)
+    assert n_cpu == n_gpu
+    assert torch.equal(c_cpu, c_gpu.cpu())
+    assert check_coloring(dgg, Comm(dev), c_gpu, exclude_color=n_gpu - 1) == 0
