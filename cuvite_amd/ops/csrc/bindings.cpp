@@ -63,6 +63,9 @@ int hub_argmax_splits();
 template <typename W>
 void launch_apply_deltas(const int64_t*, const int64_t*, const W*, int64_t,
                          int64_t, int64_t, int64_t*, W*, hipStream_t);
+template <typename W>
+void launch_recount(const int64_t*, const W*, int64_t, int64_t, int64_t*,
+                    W*, hipStream_t);
 void launch_coloring_minmax(const int64_t*, const int32_t*, const int64_t*,
                             const bool*, const int64_t*, int64_t, int64_t,
                             const int64_t*, int, int64_t*, int64_t*,
@@ -294,6 +297,28 @@ std::vector<at::Tensor> hub_moves(
   return {target_hub, cw_hub};
 }
 
+// Fresh world-1 recount of the community aggregates (see recount_kernel).
+void recount_(at::Tensor labels, at::Tensor v_degree, int64_t base,
+              at::Tensor size, at::Tensor degree) {
+  CHECK_DEV(labels); CHECK_CONT(labels);
+  CHECK_DEV(v_degree); CHECK_CONT(v_degree);
+  CHECK_DEV(size); CHECK_CONT(size);
+  CHECK_DEV(degree); CHECK_CONT(degree);
+  TORCH_CHECK(labels.scalar_type() == at::kLong);
+  TORCH_CHECK(size.scalar_type() == at::kLong);
+  TORCH_CHECK(labels.numel() == v_degree.numel());
+  size.zero_();
+  degree.zero_();
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES(degree.scalar_type(), "recount", [&] {
+    cuvite::launch_recount<scalar_t>(
+        labels.data_ptr<int64_t>(), v_degree.data_ptr<scalar_t>(),
+        labels.numel(), base, size.data_ptr<int64_t>(),
+        degree.data_ptr<scalar_t>(), stream);
+  });
+  C10_HIP_CHECK(hipGetLastError());
+}
+
 // Coarsening aggregate: sort packed (src,dst) keys with weights (narrow-bit
 // radix) and sum duplicate keys. Returns (uniq int64 [n], sums W [n],
 // count int32[1]) — caller trims to count (ref fill_newEdgesMap +
@@ -431,6 +456,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "(rocPRIM)");
   m.def("coloring_minmax", &coloring_minmax,
         "per-vertex multi-hash min/max over competing neighbors (HIP)");
+  m.def("recount_", &recount_,
+        "fresh community size/degree recount from labels (HIP)");
   m.def("hub_moves", &hub_moves,
         "full device-side hub move: segsort + reduce_by_key + argmax");
 }

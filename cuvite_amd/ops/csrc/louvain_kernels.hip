@@ -804,6 +804,38 @@ template void launch_scatter_add<float>(float*, const int64_t*, const float*,
 template void launch_scatter_add<double>(double*, const int64_t*,
                                          const double*, int64_t, hipStream_t);
 
+// Fresh recount of the community aggregates from scratch: one pass, two
+// atomics per vertex. At world=1 in a heavy-move regime this beats the
+// delta update (4 atomics per MOVED vertex; rocprof: 25 ms vs ~12 at s26
+// where most vertices move every oscillating sweep). The distributed path
+// keeps deltas (owners need only the cross-rank changes).
+template <typename W>
+__global__ void recount_kernel(const int64_t* __restrict__ labels,
+                               const W* __restrict__ v_degree, int64_t nv,
+                               int64_t base, int64_t* __restrict__ size,
+                               W* __restrict__ degree) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+       i += stride) {
+    const int64_t c = labels[i] - base;
+    atomicAdd((unsigned long long*)&size[c], 1ull);
+    unsafeAtomicAdd(&degree[c], v_degree[i]);
+  }
+}
+
+template <typename W>
+void launch_recount(const int64_t* labels, const W* v_degree, int64_t nv,
+                    int64_t base, int64_t* size, W* degree,
+                    hipStream_t stream) {
+  if (nv == 0) return;
+  hipLaunchKernelGGL((recount_kernel<W>), dim3(grid_for(nv, 256)), dim3(256),
+                     0, stream, labels, v_degree, nv, base, size, degree);
+}
+template void launch_recount<float>(const int64_t*, const float*, int64_t,
+                                    int64_t, int64_t*, float*, hipStream_t);
+template void launch_recount<double>(const int64_t*, const double*, int64_t,
+                                     int64_t, int64_t*, double*, hipStream_t);
+
 template <typename W>
 void launch_apply_deltas(const int64_t* target, const int64_t* curr,
                          const W* v_degree, int64_t nv, int64_t base,

@@ -415,3 +415,21 @@ def test_coloring_gpu_matches_cpu_exact():
     assert n_cpu == n_gpu
     assert torch.equal(c_cpu, c_gpu.cpu())
     assert check_coloring(dgg, Comm(dev), c_gpu, exclude_color=n_gpu - 1) == 0
+
+
+def test_recount_matches_torch():
+    from cuvite_amd import ops
+    dev = torch.device("cuda:0")
+    torch.manual_seed(9)
+    nv, base = 40000, 500
+    labels = torch.randint(base, base + nv, (nv,), dtype=torch.int64,
+                           device=dev)
+    vdeg = torch.rand(nv, dtype=torch.float64, device=dev)
+    size = torch.randint(0, 9, (nv,), dtype=torch.int64, device=dev)
+    degree = torch.rand(nv, dtype=torch.float64, device=dev)
+    ops._require().recount_(labels, vdeg, base, size, degree)
+    size_ref = torch.bincount(labels - base, minlength=nv)
+    degree_ref = torch.zeros(nv, dtype=torch.float64, device=dev)
+    degree_ref.index_add_(0, labels - base, vdeg)
+    assert torch.equal(size, size_ref)
+    assert torch.allclose(degree, degree_ref)
