@@ -64,8 +64,8 @@ template <typename W>
 void launch_apply_deltas(const int64_t*, const int64_t*, const W*, int64_t,
                          int64_t, int64_t, int64_t*, W*, hipStream_t);
 template <typename W>
-void launch_recount(const int64_t*, const W*, int64_t, int64_t, int64_t*,
-                    W*, hipStream_t);
+void launch_recount(const int64_t*, const W*, int64_t, int64_t, int64_t,
+                    int64_t*, W*, hipStream_t);
 void launch_coloring_minmax(const int64_t*, const int32_t*, const int64_t*,
                             const bool*, const int64_t*, int64_t, int64_t,
                             const int64_t*, int, int64_t*, int64_t*,
@@ -313,7 +313,7 @@ void recount_(at::Tensor labels, at::Tensor v_degree, int64_t base,
   AT_DISPATCH_FLOATING_TYPES(degree.scalar_type(), "recount", [&] {
     cuvite::launch_recount<scalar_t>(
         labels.data_ptr<int64_t>(), v_degree.data_ptr<scalar_t>(),
-        labels.numel(), base, size.data_ptr<int64_t>(),
+        labels.numel(), base, size.numel(), size.data_ptr<int64_t>(),
         degree.data_ptr<scalar_t>(), stream);
   });
   C10_HIP_CHECK(hipGetLastError());

@@ -812,12 +812,16 @@ template void launch_scatter_add<double>(double*, const int64_t*,
 template <typename W>
 __global__ void recount_kernel(const int64_t* __restrict__ labels,
                                const W* __restrict__ v_degree, int64_t nv,
-                               int64_t base, int64_t* __restrict__ size,
+                               int64_t base, int64_t ncomm,
+                               int64_t* __restrict__ size,
                                W* __restrict__ degree) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
        i += stride) {
     const int64_t c = labels[i] - base;
+    // world-1 labels are local by construction; the bounds check is a
+    // memory-safety guard (an OOB label must not corrupt memory)
+    if (c < 0 || c >= ncomm) continue;
     atomicAdd((unsigned long long*)&size[c], 1ull);
     unsafeAtomicAdd(&degree[c], v_degree[i]);
   }
@@ -825,16 +829,19 @@ __global__ void recount_kernel(const int64_t* __restrict__ labels,
 
 template <typename W>
 void launch_recount(const int64_t* labels, const W* v_degree, int64_t nv,
-                    int64_t base, int64_t* size, W* degree,
+                    int64_t base, int64_t ncomm, int64_t* size, W* degree,
                     hipStream_t stream) {
   if (nv == 0) return;
   hipLaunchKernelGGL((recount_kernel<W>), dim3(grid_for(nv, 256)), dim3(256),
-                     0, stream, labels, v_degree, nv, base, size, degree);
+                     0, stream, labels, v_degree, nv, base, ncomm, size,
+                     degree);
 }
 template void launch_recount<float>(const int64_t*, const float*, int64_t,
-                                    int64_t, int64_t*, float*, hipStream_t);
+                                    int64_t, int64_t, int64_t*, float*,
+                                    hipStream_t);
 template void launch_recount<double>(const int64_t*, const double*, int64_t,
-                                     int64_t, int64_t*, double*, hipStream_t);
+                                     int64_t, int64_t, int64_t*, double*,
+                                     hipStream_t);
 
 template <typename W>
 void launch_apply_deltas(const int64_t* target, const int64_t* curr,
