@@ -96,7 +96,7 @@ cuvite::MoveArgs<W> make_args(const at::Tensor& rowptr, const at::Tensor& tails,
       cw.data_ptr<W>()};
 }
 
-// vlists: the 5 LDS degree-class vertex lists (hubs excluded)
+// vlists: the 7 LDS degree-class vertex lists (hubs excluded)
 // (int32, padded is NOT required; launchers pad logically by bounds checks in
 // the sub kernels via nlist).
 std::vector<at::Tensor> local_move(
