@@ -236,7 +236,8 @@ def _hub_static(inp, hubs, hdeg):
 
 
 def _hub_moves_sorted_one(inp, hubs, hdeg):
-    """Hub vertices (deg > 4096) via radix sort + segmented reduction.
+    """Hub vertices (deg > hub cut, default 6144) via radix sort +
+    segmented reduction.
     Default path: the fully-device rocPRIM pipeline (hub_moves binding).
     Fallback (CUVITE_HUB_SEGSORT=0): torch.sort of packed (hub, community)
     keys + cumsum segment sums + vectorized exact-tie-break argmax.
@@ -314,7 +315,8 @@ def local_move(inp):
     Returns (target dense comm ids [nv], cluster_weight [nv]).
 
     Degree classes 0-4 run in the LDS hash-table kernels; hub vertices
-    (deg > 4096) go through the rocPRIM segsort pipeline (_hub_moves_sorted),
+    (deg > _hub_cut()) go through the rocPRIM segsort pipeline
+    (_hub_moves_sorted),
     overlapped on separate HIP streams (disjoint vertex sets)."""
     ext = _require()
     vlists, hubs64, hdeg = _buckets_for(inp.rowptr)

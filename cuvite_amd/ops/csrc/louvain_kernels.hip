@@ -20,8 +20,10 @@
 //   class 3: deg in (256, 512]   one wave/vertex, 1024-slot LDS (48 KB/blk)
 //   class 4: deg in (512, 1024]  256-thread block/vertex, 2048-slot (24 KB)
 //   class 5: deg in (1024, 2048] 256-thread block/vertex, 4096-slot (48 KB)
-//   class 6: deg in (2048, 4096] 256-thread block/vertex, 8192-slot (96 KB)
-//   hub    : deg > 4096          hub_moves binding (rocPRIM narrow-bit
+//   class 6: deg in (2048, cut]  256-thread block/vertex, 8192-slot (96 KB;
+//                                cut = CUVITE_HUB_CUT, default 6144 after
+//                                the s26 A/B in profiles/)
+//   hub    : deg > cut (6144)    hub_moves binding (rocPRIM narrow-bit
 //                                segmented radix sort + reduce_by_key +
 //                                split-block argmax) — 2.9x faster than
 //                                the torch global-sort fallback at s26
