@@ -73,6 +73,19 @@ def main():
     if device.type == "cuda":
         torch.cuda.synchronize()
     _p(f"graph built: nv_local={dg.nv} ne_local={dg.ne}")
+    vertex_order = "natural"
+    if (comm.world == 1 and device.type == "cuda"
+            and os.environ.get("CUVITE_DEGSORT", "0") == "1"):
+        # isomorphic degree-descending relabeling: hub labels pack into a
+        # small hot region so the per-edge curr_comm gathers hit the XCD L2s
+        # (graph unchanged up to isomorphism; see generators.degree_sort_graph)
+        from cuvite_amd.generators import degree_sort_graph
+        from cuvite_amd.graph import DistGraph
+        g2, _ = degree_sort_graph(dg.g)
+        dg = DistGraph(g2, dg.partition, 0)
+        torch.cuda.synchronize()
+        vertex_order = "degree"
+        _p("degree-sorted relabeling applied")
     ne_global = float(comm.allreduce_scalar(float(dg.ne)))
     t_gen = time.perf_counter() - t_gen0
 
@@ -209,6 +222,7 @@ def main():
                 "gen_seconds": round(t_gen, 2),
                 "parallelism": f"graph1d-p{comm.world}",
                 "backend": args.backend,
+                "vertex_order": vertex_order,
             },
         }
         if conv is not None:

@@ -520,3 +520,54 @@ def karate_graph(weight_dtype: torch.dtype = torch.float64) -> Graph:
     dst = torch.cat([e[:, 1], e[:, 0]])
     w = torch.ones(src.numel(), dtype=weight_dtype)
     return Graph.from_edge_tuples(34, src, dst, w)
+
+
+def degree_sort_graph(g, chunk: int = 1 << 28):
+    """Isomorphic relabeling of a single-rank Graph with vertices ordered by
+    degree (descending): vertex new-id p = old vertex order[p]. Returns
+    (new Graph, order) where order maps new ids back to old ids.
+
+    Locality optimization for power-law graphs: the local-move kernels'
+    dominant traffic is the data-dependent curr_comm[tail] gather (PMC: ~80%
+    SQ_WAIT_ANY, profiles/round2_kernel_stats.md); hubs receive a large
+    fraction of all edges, so packing them at the front of the id space
+    turns those gathers into hits on a few hundred KB that stay resident in
+    the per-XCD L2s. The graph is unchanged up to isomorphism — same
+    degrees, weights, modularity landscape; community labels map through
+    `order`. Chunked: gathers over >INT_MAX edges overflow torch indexing."""
+    from .graph import Graph
+    dev = g.device
+    nv = g.nv
+    deg = g.degrees()
+    # stable descending degree order (ties keep original id order)
+    order = torch.argsort(-deg, stable=True)
+    inv = torch.empty(nv, dtype=torch.int64, device=dev)
+    inv[order] = torch.arange(nv, device=dev)
+    new_deg = deg[order]
+    rowptr2 = torch.zeros(nv + 1, dtype=torch.int64, device=dev)
+    rowptr2[1:] = torch.cumsum(new_deg, dim=0)
+    ne = g.ne
+    tails2 = torch.empty(ne, dtype=torch.int64, device=dev)
+    weights2 = torch.empty(ne, dtype=g.weights.dtype, device=dev)
+    old_start = g.rowptr[order]  # old CSR start per NEW row
+    # process new rows in slabs whose edge totals stay under the cap
+    row_lo = 0
+    while row_lo < nv:
+        row_hi = int(torch.searchsorted(
+            rowptr2, torch.tensor(int(rowptr2[row_lo]) + chunk,
+                                  device=dev)))
+        row_hi = max(row_lo + 1, min(row_hi, nv))
+        e0 = int(rowptr2[row_lo])
+        e1 = int(rowptr2[row_hi])
+        n = e1 - e0
+        if n:
+            seg = torch.searchsorted(
+                rowptr2, torch.arange(e0, e1, device=dev), right=True) - 1
+            eidx = old_start[seg] + \
+                (torch.arange(e0, e1, device=dev) - rowptr2[seg])
+            del seg
+            tails2[e0:e1] = inv[g.tails[eidx]]
+            weights2[e0:e1] = g.weights[eidx]
+            del eidx
+        row_lo = row_hi
+    return Graph(rowptr2, tails2, weights2), order

@@ -164,3 +164,32 @@ def test_fold_keys_range_partition_matches_flat():
         C._FOLD_CAP = old
     assert torch.equal(got[0], ref[0])
     assert torch.allclose(got[1], ref[1])
+
+
+def test_degree_sort_graph_isomorphic():
+    """Degree-sorted relabeling preserves the graph up to isomorphism and
+    Louvain lands on the same community structure (same Q on unit weights)."""
+    from cuvite_amd.generators import degree_sort_graph, rmat_graph
+    from cuvite_amd.graph import single_partition
+    from cuvite_amd.louvain import louvain, LouvainConfig
+    from cuvite_amd.parallel import Comm
+    g = rmat_graph(9, 12, seed=4)
+    g2, order = degree_sort_graph(g)
+    assert torch.equal(g2.degrees(), g.degrees()[order])
+    # degrees non-increasing
+    d = g2.degrees()
+    assert bool((d[:-1] >= d[1:]).all())
+    # per-row multisets map through the permutation
+    inv = torch.empty(g.nv, dtype=torch.int64)
+    inv[order] = torch.arange(g.nv)
+    for p in range(0, g.nv, 37):
+        v = int(order[p])
+        a = sorted(zip(inv[g.tails[g.rowptr[v]:g.rowptr[v + 1]]].tolist(),
+                       g.weights[g.rowptr[v]:g.rowptr[v + 1]].tolist()))
+        b = sorted(zip(g2.tails[g2.rowptr[p]:g2.rowptr[p + 1]].tolist(),
+                       g2.weights[g2.rowptr[p]:g2.rowptr[p + 1]].tolist()))
+        assert a == b
+    # same modularity reached (trajectories differ only via id tie-breaks)
+    r1 = louvain(single_partition(g), Comm(), LouvainConfig(backend="torch"))
+    r2 = louvain(single_partition(g2), Comm(), LouvainConfig(backend="torch"))
+    assert abs(r1.modularity - r2.modularity) < 0.02
