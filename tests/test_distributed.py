@@ -333,3 +333,28 @@ def test_per_iteration_comm_rounds():
     assert c["all_reduce"] == 1, c          # modularity only
     assert c["all_gather"] <= 2, c          # count negotiations
     assert c["batch"] <= 4, c               # label xchg + info req/rep + deltas
+
+
+def _w_degsort(rank, world):
+    from cuvite_amd.generators import degree_sort_dist, rmat_graph
+    g = rmat_graph(9, 12, seed=4)
+    dg = _shard(g, rank, world)
+    comm = Comm()
+    dg2 = degree_sort_dist(dg, comm)
+    res = louvain(dg2, comm, LouvainConfig(backend="torch"))
+    return (dg2.g.degrees().cpu(), dg2.g.rowptr[-1].item(), res.modularity,
+            float(dg2.g.weights.sum()))
+
+
+def test_degree_sort_dist_preserves_graph():
+    """Per-rank degree relabeling (bench preprocessing): degrees become
+    non-increasing within each rank's range, edge count and total weight are
+    preserved, and Louvain reaches comparable modularity."""
+    g = rmat_graph(9, 12, seed=4)
+    ref = louvain(single_partition(g), Comm(), LouvainConfig(backend="torch"))
+    outs = run_dist(2, _w_degsort)
+    for deg, ne_local, q, wsum in outs:
+        assert bool((deg[:-1] >= deg[1:]).all())
+        assert abs(q - ref.modularity) < 0.03
+    assert sum(o[1] for o in outs) == g.ne
+    assert sum(o[3] for o in outs) == pytest.approx(float(g.weights.sum()))
