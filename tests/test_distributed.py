@@ -346,13 +346,15 @@ def _w_degsort(rank, world):
             float(dg2.g.weights.sum()))
 
 
-def test_degree_sort_dist_preserves_graph():
+@pytest.mark.parametrize("world", [2, 3])
+def test_degree_sort_dist_preserves_graph(world):
     """Per-rank degree relabeling (bench preprocessing): degrees become
     non-increasing within each rank's range, edge count and total weight are
-    preserved, and Louvain reaches comparable modularity."""
+    preserved, and Louvain reaches comparable modularity. world=3 exercises
+    the unequal-shard padding in the id-map allgather."""
     g = rmat_graph(9, 12, seed=4)
     ref = louvain(single_partition(g), Comm(), LouvainConfig(backend="torch"))
-    outs = run_dist(2, _w_degsort)
+    outs = run_dist(world, _w_degsort)
     for deg, ne_local, q, wsum in outs:
         assert bool((deg[:-1] >= deg[1:]).all())
         assert abs(q - ref.modularity) < 0.03
