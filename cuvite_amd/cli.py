@@ -88,6 +88,12 @@ def build_parser() -> argparse.ArgumentParser:
                     default="auto")
     ap.add_argument("--backend", choices=["auto", "hip", "torch"],
                     default="auto", help="local-move backend")
+    ap.add_argument("--vertex-order", choices=["natural", "degree"],
+                    default="natural",
+                    help="degree: isomorphic per-rank degree-descending "
+                         "relabeling before clustering (hub-label locality, "
+                         "+23%% at R-MAT s26 on MI355X); outputs are mapped "
+                         "back to the input vertex order")
     ap.add_argument("--threshold", type=float, default=1.0e-6)
     ap.add_argument("--max-phases", type=int, default=0,
                     help="cap the number of phases (0 = reference default)")
@@ -212,6 +218,15 @@ def main(argv=None) -> int:
     if args.just_process:
         return 0
 
+    vo_inv = None     # new local pos -> old local pos (inverse applied later)
+    vo_gmap = None    # old gid -> new gid
+    if args.vertex_order == "degree":
+        from .generators import degree_sort_dist
+        dg, vo_order, vo_gmap = degree_sort_dist(dg, comm, return_maps=True)
+        vo_inv = torch.empty_like(vo_order)
+        vo_inv[vo_order] = torch.arange(vo_order.numel(),
+                                        device=vo_order.device)
+
     cfg = LouvainConfig(
         threshold=args.threshold,
         threshold_scaling=args.threshold_cycling,
@@ -259,8 +274,18 @@ def main(argv=None) -> int:
 
     builtin_truth = getattr(args, "_lfr_truth", None)
     if args.output or args.ground_truth or builtin_truth is not None:
+        final_comm = res.communities
+        if vo_inv is not None:
+            # map back to the INPUT vertex order: original local vertex v sits
+            # at relabeled position vo_inv[v]; label values (new gids) map
+            # through the global inverse so dumps match the input id space
+            final_comm = final_comm[vo_inv]
+            inv_gmap = torch.empty_like(vo_gmap)
+            inv_gmap[vo_gmap] = torch.arange(vo_gmap.numel(),
+                                             device=vo_gmap.device)
+            final_comm = inv_gmap[final_comm]
         # gather final communities to root in vertex order
-        allc = comm.gather_cat(res.communities.to(comm.device), root=0)
+        allc = comm.gather_cat(final_comm.to(comm.device), root=0)
         if comm.rank == 0:
             allc = allc.cpu()
             if args.output:

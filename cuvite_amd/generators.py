@@ -560,7 +560,7 @@ def _reorder_rows(g, order, inv_or_map, chunk: int = 1 << 28):
     return Graph(rowptr2, tails2, weights2)
 
 
-def degree_sort_dist(dg, comm):
+def degree_sort_dist(dg, comm, return_maps: bool = False):
     """Per-rank-range degree-descending relabeling of a DistGraph: each rank
     permutes vertices WITHIN its own contiguous [base, bound) range (owners,
     partition and edge balance unchanged), so new gid of old vertex v =
@@ -594,7 +594,12 @@ def degree_sort_dist(dg, comm):
         gmap = torch.cat([outs[p][:dg.partition.nv_local(p)]
                           for p in range(comm.world)])
     g2 = _reorder_rows(dg.g, order, gmap)
-    return DistGraph(g2, dg.partition, dg.rank)
+    dg2 = DistGraph(g2, dg.partition, dg.rank)
+    if return_maps:
+        # order: new local position -> old local position;
+        # gmap: old global id -> new global id
+        return dg2, order, gmap
+    return dg2
 
 
 def degree_sort_graph(g, chunk: int = 1 << 28):
