@@ -16,7 +16,6 @@ Conventions:
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Optional
 
 import torch
 
