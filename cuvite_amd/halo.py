@@ -122,44 +122,6 @@ def exchange_ghost_labels(ctx: HaloContext, curr_comm: torch.Tensor) -> torch.Te
     return out
 
 
-def fetch_remote_comm_info(ctx: HaloContext, remote_gids: torch.Tensor,
-                           local_size: torch.Tensor, local_degree: torch.Tensor):
-    """Fetch (size, degree) of remote communities from their owners.
-    remote_gids: sorted unique int64 global community ids NOT owned here.
-    Returns (sizes int64 [nrc], degrees W [nrc]) aligned with remote_gids.
-    Ref: rounds 2-3 of fillRemoteCommunities (louvain.cpp:2688-2959)."""
-    comm, dg = ctx.comm, ctx.dg
-    dev = remote_gids.device
-    W = local_degree.dtype
-    if comm.world == 1:
-        assert remote_gids.numel() == 0, "remote comms with world=1"
-        return (torch.empty(0, dtype=torch.int64, device=dev),
-                torch.empty(0, dtype=W, device=dev))
-    # NOTE: even with no remote references of our own we must participate —
-    # peers may be requesting OUR community info (collective protocol).
-    parts = dg.partition.parts.to(dev)
-    offs = torch.searchsorted(remote_gids, parts)
-    reqs = [remote_gids[offs[p]:offs[p + 1]] for p in range(comm.world)]
-    got = comm.all_to_all_v(reqs)
-    # one fused reply per peer: [size-bits, degree-bits] as fp64 payload
-    # (sizes are exact in fp64 up to 2^53; halves the p2p rounds over xGMI)
-    reply = []
-    for p in range(comm.world):
-        if p == comm.rank or got[p].numel() == 0:
-            reply.append(torch.empty(0, dtype=torch.float64, device=dev))
-            continue
-        li = got[p] - dg.base
-        reply.append(torch.cat([local_size[li].to(torch.float64),
-                                local_degree[li].to(torch.float64)]))
-    req_counts = [2 * int(r.numel()) for r in reqs]
-    back = comm.all_to_all_v(reply, recv_counts=req_counts)
-    sizes = torch.cat([back[p][:back[p].numel() // 2]
-                       for p in range(comm.world)]).to(torch.int64)
-    degrees = torch.cat([back[p][back[p].numel() // 2:]
-                         for p in range(comm.world)]).to(W)
-    return sizes, degrees
-
-
 def fetch_comm_info_lists(ctx: HaloContext, reqs: List[torch.Tensor],
                           local_size: torch.Tensor,
                           local_degree: torch.Tensor):
@@ -167,10 +129,9 @@ def fetch_comm_info_lists(ctx: HaloContext, reqs: List[torch.Tensor],
     requests already grouped per owner rank (reqs[p] = gids owned by p, any
     order). Returns per-peer lists (sizes[p] int64, degrees[p] W) aligned
     with reqs[p]. Collective: every rank must call with world-length lists.
-    Ref: rounds 2-3 of fillRemoteCommunities (louvain.cpp:2688-2959); unlike
-    fetch_remote_comm_info this variant imposes no global-sort requirement
-    on the request set (the phase-persistent universe in louvain.densify is
-    append-ordered)."""
+    Ref: rounds 2-3 of fillRemoteCommunities (louvain.cpp:2688-2959). No
+    global-sort requirement on the request set (the phase-persistent
+    universe in louvain.densify is append-ordered)."""
     comm, dg = ctx.comm, ctx.dg
     W = local_degree.dtype
     got = comm.all_to_all_v(reqs)
