@@ -10,8 +10,9 @@ MI355X-native redesign: everything is device tensors end to end. The ghost
 set is static within a phase, so the label exchange runs with pre-negotiated
 sizes into persistent device buffers (grouped RCCL p2p over xGMI). The dense
 remote-community remap, which the reference rebuilds on the HOST every
-iteration (louvain_cuda.cu:2260-2378), is computed on-device with
-sort/unique/searchsorted.
+iteration (louvain_cuda.cu:2260-2378), lives in a phase-persistent
+remote-community universe on device (louvain.PhaseState.densify) — only
+CHANGED labels are remapped per iteration, with no sort/unique in the loop.
 """
 
 from __future__ import annotations
