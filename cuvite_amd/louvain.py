@@ -10,8 +10,10 @@ vertex-ordered moves (ref louvain.cpp:756-2101), threshold cycling
 
 All state lives on the compute device; per-iteration host work is zero on the
 GPU path (the reference rebuilds its dense community remap on the host every
-iteration, louvain_cuda.cu:2260-2378 — here it is torch sort/unique/
-searchsorted on device).
+iteration, louvain_cuda.cu:2260-2378 — here remote community ids are interned
+once into a phase-persistent on-device universe and only CHANGED labels are
+remapped per iteration, with no sort/unique in the loop; see
+PhaseState.densify).
 """
 
 from __future__ import annotations
